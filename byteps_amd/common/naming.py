@@ -1,0 +1,125 @@
+"""Tensor naming and key assignment.
+
+Reproduces the reference's naming / key-space layout:
+
+- canonical tensor names ``byteps.Gradient.<param>``, ``byteps.Parameter.<n>``,
+  ``byteps.AsyncParam.<n>`` (reference torch/ops.cc:37-43,104);
+- declared keys assigned in sorted-declaration order so all ranks agree
+  without communication (reference common/global.cc:412-429);
+- partition keys ``declared_key << 16 | part`` — up to 2^16 partitions per
+  tensor (reference common/operations.cc:306-311);
+- server assignment by djb2 hash of the partition key with per-server load
+  accounting (reference common/global.cc:628-677).
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List
+
+PART_BITS = 16
+MAX_PARTS = 1 << PART_BITS
+
+
+def gradient_name(param_name: str) -> str:
+    return "byteps.Gradient." + param_name
+
+
+def parameter_name(name: str) -> str:
+    return "byteps.Parameter." + name
+
+
+def async_param_name(name: str) -> str:
+    return "byteps.AsyncParam." + name
+
+
+def partition_key(declared_key: int, part: int) -> int:
+    assert 0 <= part < MAX_PARTS
+    return (declared_key << PART_BITS) | part
+
+
+def declared_key_of(pkey: int) -> int:
+    return pkey >> PART_BITS
+
+
+def part_of(pkey: int) -> int:
+    return pkey & (MAX_PARTS - 1)
+
+
+def djb2(key: int) -> int:
+    """djb2 over the decimal string of the key — the reference's default
+    server-assignment hash (reference common/global.cc:628-677)."""
+    h = 5381
+    for ch in str(key):
+        h = ((h * 33) + ord(ch)) & 0xFFFFFFFFFFFFFFFF
+    return h
+
+
+class NameRegistry:
+    """Deterministic name → declared-key table.
+
+    Keys are assigned incrementally in declaration-arrival order
+    (reference common/global.cc:412-429); callers must declare in the same
+    deterministic order on every worker (the reference's Python plugins
+    declare in sorted passes for exactly this reason,
+    torch/__init__.py:95-100 — our engine declares in model registration
+    order, identical across ranks).  Once assigned, a key never changes —
+    elastic resume re-declares in original order to keep keys stable
+    (reference common/global.cc:431-436).
+    """
+
+    def __init__(self) -> None:
+        self._keys: Dict[str, int] = {}
+        self._declare_order: List[str] = []
+
+    def declare(self, name: str) -> int:
+        if name not in self._keys:
+            self._keys[name] = len(self._declare_order)
+            self._declare_order.append(name)
+        return self._keys[name]
+
+    def key(self, name: str) -> int:
+        return self._keys[name]
+
+    def __contains__(self, name: str) -> bool:
+        return name in self._keys
+
+    def __len__(self) -> int:
+        return len(self._keys)
+
+    @property
+    def declare_order(self) -> List[str]:
+        return list(self._declare_order)
+
+    def redeclare_all(self) -> None:
+        """Elastic resume: re-declare every known tensor in original order
+        (reference common/operations.cc:96-119)."""
+        order = self._declare_order
+        self._keys, self._declare_order = {}, []
+        for n in order:
+            self.declare(n)
+
+
+class ServerAssigner:
+    """Key → server with djb2 hash + greedy load balancing.
+
+    The reference hashes first and falls back to accumulated-load accounting
+    (common/global.cc:660-667).  We hash to a preferred server and accept it
+    unless its accumulated bytes exceed the least-loaded server by more than
+    one partition, in which case the least-loaded server takes the key —
+    deterministic given identical declaration sequences on all workers.
+    """
+
+    def __init__(self, num_servers: int) -> None:
+        self.num_servers = max(1, num_servers)
+        self.load = [0] * self.num_servers
+        self.table: Dict[int, int] = {}
+
+    def assign(self, pkey: int, nbytes: int) -> int:
+        if pkey in self.table:
+            return self.table[pkey]
+        pref = djb2(pkey) % self.num_servers
+        least = min(range(self.num_servers), key=lambda s: self.load[s])
+        srv = pref if self.load[pref] - self.load[least] <= nbytes else least
+        self.table[pkey] = srv
+        self.load[srv] += nbytes
+        return srv

@@ -1,0 +1,341 @@
+"""The gradient-synchronization engine.
+
+MI355X-native re-design of the reference's 12-stage queue pipeline
+(reference common/core_loops.cc:755-855, common/scheduled_queue.cc).  One
+process per GPU; the intra-node data plane is RCCL over xGMI driven through
+``torch.distributed`` (gloo on CPU for tests).  Design:
+
+- **Persistent flat buckets**: parameters are packed (in reverse
+  registration order, so the gradients backward produces first complete
+  first) into flat buffers; ``param.grad`` is a zero-copy *view* into its
+  bucket (the reference instead copied every gradient into pinned shared
+  memory, common/core_loops.cc:378-443 — on MI355X gradients stay resident
+  in HBM3E and PCIe is only crossed in PS mode, with compressed bytes).
+- **Priority scheduling**: ready buckets are drained through a max-heap
+  keyed by (priority, -index) (reference scheduled_queue.cc:86-95); byte
+  credits (``BPS_SCHEDULING_CREDIT``) bound in-flight communication
+  (reference scheduled_queue.cc:33-45).
+- **Overlap**: collectives are issued with ``async_op=True`` → RCCL runs on
+  its own HIP stream, overlapping the remaining backward; ``synchronize()``
+  waits on the Work events (and in PS mode the KV completion flags).
+- **PS mode** (``BPS_NUM_SERVER>0`` or ``BPS_FORCE_DISTRIBUTED``): per
+  bucket reduce-scatter → (compress →) D2H into pinned staging → push/pull
+  to the C++ KV server → H2D (→ decompress) → all-gather, with only this
+  rank's shard crossing PCIe (reference route construction,
+  common/operations.cc:429-485).
+"""
+
+from __future__ import annotations
+
+import heapq
+import threading
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+from .. import common as C
+from ..common.config import Config
+from ..common.logging_util import get_logger
+from ..common.partition import PartitionPlan, plan_partitions
+
+log = get_logger()
+
+
+# --------------------------------------------------------------------------
+# Bucket: one schedulable flat buffer
+# --------------------------------------------------------------------------
+
+@dataclass
+class Bucket:
+    plan: PartitionPlan
+    buffer: torch.Tensor                      # flat grad buffer (persistent)
+    params: List[torch.nn.Parameter]          # params whose grads live here
+    grads: List[torch.Tensor]                 # the views (same order)
+    priority: int = 0
+    declared_key: int = 0
+    # per-step state
+    ready_count: int = 0
+    issued: bool = False
+    work: Optional[object] = None             # dist Work handle
+    ps_ticket: Optional[object] = None        # PS pipeline ticket
+    done_event: Optional[torch.cuda.Event] = None
+
+    @property
+    def nbytes(self) -> int:
+        return self.buffer.numel() * self.buffer.element_size()
+
+    def reset(self) -> None:
+        self.ready_count = 0
+        self.issued = False
+        self.work = None
+        self.ps_ticket = None
+        self.done_event = None
+
+
+# --------------------------------------------------------------------------
+# Engine
+# --------------------------------------------------------------------------
+
+class GradEngine:
+    """Owns the buckets for one optimizer/DDP instance.
+
+    ``params`` must be passed in registration (model) order on every rank —
+    bucket layout is a pure function of that order, so all ranks agree
+    without communication (mirrors the reference's sorted declare,
+    common/global.cc:412-429).
+    """
+
+    def __init__(
+        self,
+        named_params: Sequence,               # iterable of (name, Parameter)
+        process_group: Optional[dist.ProcessGroup] = None,
+        partition_bytes: Optional[int] = None,
+        average: bool = True,
+        grad_dtype: Optional[torch.dtype] = None,
+        prescale: bool = False,
+    ) -> None:
+        C._require_init()
+        self.cfg: Config = C.get_config()
+        self.group = process_group
+        self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self.average = average
+        self.prescale = prescale and self.world > 1
+        self._lock = threading.Lock()
+        self._pending: List = []              # heap of (-priority, idx)
+        self._inflight_bytes = 0
+        self._credit = self.cfg.scheduling_credit  # 0 → unlimited
+        self._deferred: List[int] = []
+        self._hook_handles: List = []
+        self._sync_enabled = True
+        self._step = 0
+        self._ready_params = 0
+        # invoked (from the last autograd hook) once every param's grad is
+        # ready — DDP uses it to self-synchronize (reference
+        # parallel/distributed.py:261-270)
+        self.on_all_ready: Optional[Callable] = None
+
+        named = [(n, p) for n, p in named_params if p.requires_grad]
+        if not named:
+            raise ValueError("GradEngine: no parameters require grad")
+        self.param_names = [n for n, _ in named]
+        self.params = [p for _, p in named]
+        for name in self.param_names:
+            C._state.registry.declare("byteps.Gradient." + name)
+
+        dt = grad_dtype or self.params[0].dtype
+        dev = self.params[0].device
+        part_bytes = partition_bytes or self.cfg.partition_bytes
+        part_elems = max(part_bytes // torch.tensor([], dtype=dt).element_size(), 4096)
+
+        # Bucket in REVERSE registration order: backward runs output → input,
+        # so the last-registered params produce gradients first and land in
+        # bucket 0 (highest priority) — the priority semantics of the
+        # reference (mxnet/__init__.py:58-60, scheduled_queue.cc:86-95).
+        rev = list(range(len(self.params)))[::-1]
+        sizes = [self.params[i].numel() for i in rev]
+        # align so reduce-scatter shards divide evenly for any world ≤ 64
+        plans = plan_partitions(sizes, part_elems, align=64)
+
+        self.buckets: List[Bucket] = []
+        self.param_bucket: Dict[int, List[Bucket]] = {}   # param idx → buckets
+        self._param_span_count: Dict[int, int] = {i: 0 for i in range(len(self.params))}
+        for plan in plans:
+            buf = torch.zeros(plan.numel, dtype=dt, device=dev)
+            bucket = Bucket(plan=plan, buffer=buf, params=[], grads=[],
+                            priority=plan.priority)
+            key = C._state.registry.declare(
+                "byteps.Partition.%d" % plan.index)
+            bucket.declared_key = key
+            for span in plan.spans:
+                pidx = rev[span.param_index]
+                p = self.params[pidx]
+                view = buf.narrow(0, span.offset, span.numel)
+                bucket.params.append(p)
+                bucket.grads.append(view)
+                self.param_bucket.setdefault(pidx, []).append(bucket)
+                self._param_span_count[pidx] += 1
+            self.buckets.append(bucket)
+
+        # Attach p.grad views.  A param split across buckets cannot be a
+        # single view — those (rare: only params > partition size) get a
+        # private grad and a copy at ready time.
+        self._split_params: Dict[int, List] = {}
+        for pidx, p in enumerate(self.params):
+            bks = self.param_bucket[pidx]
+            if len(bks) == 1 and self._single_span(bks[0], p) is not None:
+                view = self._single_span(bks[0], p)
+                p.grad = view.view_as(p)
+            else:
+                # split across partitions: keep torch-allocated grad, copy
+                # into the views when the grad is produced
+                p.grad = torch.zeros_like(p)
+                self._split_params[pidx] = bks
+
+        self._ps = None
+        if C._state.ps_enabled:
+            from . import ps_pipeline
+            self._ps = ps_pipeline.get_pipeline(self)
+
+        self._attach_hooks()
+        self._expected_ready = {id(b): len(b.params) for b in self.buckets}
+        log.debug("GradEngine: %d params → %d buckets (%.1f MiB each max)",
+                  len(self.params), len(self.buckets),
+                  part_elems * buf.element_size() / 2**20)
+
+    # -- helpers -----------------------------------------------------------
+
+    def _single_span(self, bucket: Bucket, p: torch.nn.Parameter):
+        spans = [g for q, g in zip(bucket.params, bucket.grads) if q is p]
+        if len(spans) == 1 and spans[0].numel() == p.numel():
+            return spans[0]
+        return None
+
+    def _attach_hooks(self) -> None:
+        for pidx, p in enumerate(self.params):
+            h = p.register_post_accumulate_grad_hook(
+                self._make_hook(pidx))
+            self._hook_handles.append(h)
+
+    def _make_hook(self, pidx: int) -> Callable:
+        def hook(param: torch.nn.Parameter) -> None:
+            self._on_grad_ready(pidx)
+        return hook
+
+    # -- per-step flow ------------------------------------------------------
+
+    def _on_grad_ready(self, pidx: int) -> None:
+        if not self._sync_enabled:
+            return
+        p = self.params[pidx]
+        if pidx in self._split_params:
+            # copy the private grad into its bucket views
+            flat = p.grad.reshape(-1)
+            off = 0
+            for b in self._split_params[pidx]:
+                for q, g in zip(b.params, b.grads):
+                    if q is p:
+                        g.copy_(flat.narrow(0, off, g.numel()))
+                        off += g.numel()
+        with self._lock:
+            self._ready_params += 1
+            all_ready = self._ready_params == len(self.params)
+            for b in self.param_bucket[pidx]:
+                b.ready_count += 1
+                if b.ready_count == len(b.params) and not b.issued:
+                    heapq.heappush(self._pending, (-b.priority, b.plan.index))
+            self._drain_locked()
+        if all_ready and self.on_all_ready is not None:
+            self.on_all_ready()
+
+    def _drain_locked(self) -> None:
+        """Issue every currently-ready bucket in priority order, subject to
+        the byte credit."""
+        while self._pending:
+            negp, idx = self._pending[0]
+            b = self.buckets[idx]
+            if self._credit and self._inflight_bytes + b.nbytes > self._credit \
+                    and self._inflight_bytes > 0:
+                break
+            heapq.heappop(self._pending)
+            self._issue(b)
+
+    def _issue(self, b: Bucket) -> None:
+        if b.issued:
+            return
+        b.issued = True
+        self._inflight_bytes += b.nbytes
+        if C._state.tracer is not None:
+            C._state.tracer.begin(b.declared_key, "comm", self._step)
+        if self.world <= 1 and self._ps is None:
+            b.work = None
+            return
+        if self.prescale:
+            b.buffer.div_(self.world)
+        if self._ps is not None:
+            b.ps_ticket = self._ps.submit(b)
+        elif self.world > 1:
+            b.work = dist.all_reduce(
+                b.buffer, op=dist.ReduceOp.SUM, group=self.group,
+                async_op=True)
+
+    # -- public API ---------------------------------------------------------
+
+    def zero_grad(self) -> None:
+        """Zero all bucket buffers (and private split grads)."""
+        torch._foreach_zero_([b.buffer for b in self.buckets])
+        for pidx in self._split_params:
+            self.params[pidx].grad.zero_()
+
+    def set_sync_enabled(self, enabled: bool) -> None:
+        """Gradient accumulation: when disabled, hooks accumulate into the
+        flat buffers without communicating (reference no_sync,
+        parallel/distributed.py:184-207)."""
+        self._sync_enabled = enabled
+
+    def flush(self) -> None:
+        """Force-issue buckets whose params produced no grad this step
+        (their spans hold zeros — contributing zero to the sum is correct)."""
+        with self._lock:
+            for b in self.buckets:
+                if not b.issued:
+                    heapq.heappush(self._pending, (-b.priority, b.plan.index))
+            self._credit_save, self._credit = getattr(self, "_credit", 0), 0
+            self._drain_locked()
+            self._credit = self._credit_save
+
+    def synchronize(self) -> None:
+        """Wait for all issued buckets; apply deferred averaging; reset."""
+        if not self._sync_enabled:
+            return
+        self.flush()
+        for b in self.buckets:
+            if b.ps_ticket is not None:
+                self._ps.wait(b.ps_ticket)
+            elif b.work is not None:
+                b.work.wait()
+        if self.average and self.world > 1 and not self.prescale:
+            torch._foreach_div_([b.buffer for b in self.buckets],
+                                float(self.world))
+        if self.average and self._split_params:
+            # split params read back their averaged grads
+            for pidx, bks in self._split_params.items():
+                p = self.params[pidx]
+                flat = p.grad.reshape(-1)
+                off = 0
+                for b in bks:
+                    for q, g in zip(b.params, b.grads):
+                        if q is p:
+                            flat.narrow(0, off, g.numel()).copy_(g)
+                            off += g.numel()
+        if C._state.tracer is not None:
+            for b in self.buckets:
+                C._state.tracer.end(b.declared_key, "comm", self._step)
+        for b in self.buckets:
+            b.reset()
+        self._ready_params = 0
+        self._inflight_bytes = 0
+        self._step += 1
+
+    def detach(self) -> None:
+        for h in self._hook_handles:
+            h.remove()
+        self._hook_handles.clear()
+
+
+_engines: List[GradEngine] = []
+
+
+def register_engine(e: GradEngine) -> GradEngine:
+    _engines.append(e)
+    return e
+
+
+def _shutdown_engine() -> None:
+    for e in _engines:
+        try:
+            e.detach()
+        except Exception:
+            pass
+    _engines.clear()
