@@ -1,0 +1,221 @@
+#!/usr/bin/env python3
+"""byteps_amd flagship benchmark — images/sec for ResNet-50 (default) or
+BERT-large data-parallel training on MI355X, synthetic data, random-init
+weights (metric and vehicle per BASELINE.json / reference
+example/pytorch/benchmark_byteps.py:74-130).
+
+Launched by the driver as:
+  python bench.py --gpus N --steps K --warmup W          (N=1)
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+Rank 0 prints exactly ONE JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "vgg16", "bert-large", "mlp"])
+    p.add_argument("--batch-size", type=int, default=0,
+                   help="per-GPU batch (default: 64 images / 8 seqs)")
+    p.add_argument("--seq-len", type=int, default=512)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
+                   help="hipGraph-capture the train step")
+    p.add_argument("--device", default="cuda" if torch.cuda.is_available()
+                   else "cpu")
+    p.add_argument("--compression", default="none",
+                   choices=["none", "onebit", "topk", "randomk", "dithering"])
+    p.add_argument("--partition-mb", type=int, default=0,
+                   help="override BPS_PARTITION_BYTES (MiB)")
+    return p.parse_args()
+
+
+def build(args, device):
+    from byteps_amd import models
+    if args.model == "resnet50":
+        net = models.resnet50()
+        batch = args.batch_size or 64
+        x = torch.randn(batch, 3, 224, 224, device=device)
+        y = torch.randint(0, 1000, (batch,), device=device)
+        if device.type == "cuda":
+            net = net.to(memory_format=torch.channels_last)
+            x = x.to(memory_format=torch.channels_last)
+        loss_fn = torch.nn.CrossEntropyLoss()
+
+        def step_fn(m):
+            return loss_fn(m(x), y)
+        per_step_items = batch
+        unit = "images/sec"
+    elif args.model == "vgg16":
+        net = models.vgg16()
+        batch = args.batch_size or 64
+        x = torch.randn(batch, 3, 224, 224, device=device)
+        y = torch.randint(0, 1000, (batch,), device=device)
+        if device.type == "cuda":
+            net = net.to(memory_format=torch.channels_last)
+            x = x.to(memory_format=torch.channels_last)
+        loss_fn = torch.nn.CrossEntropyLoss()
+
+        def step_fn(m):
+            return loss_fn(m(x), y)
+        per_step_items = batch
+        unit = "images/sec"
+    elif args.model == "bert-large":
+        net = models.bert_large()
+        batch = args.batch_size or 8
+        S = args.seq_len
+        ids = torch.randint(0, 30522, (batch, S), device=device)
+        labels = torch.randint(0, 30522, (batch, S), device=device)
+
+        def step_fn(m):
+            mod = m.module if hasattr(m, "module") else m
+            return mod.loss(ids, labels)
+        per_step_items = batch * S
+        unit = "tokens/sec"
+    else:  # mlp
+        net = models.mnist_mlp()
+        batch = args.batch_size or 64
+        x = torch.randn(batch, 1, 28, 28, device=device)
+        y = torch.randint(0, 10, (batch,), device=device)
+        loss_fn = torch.nn.CrossEntropyLoss()
+
+        def step_fn(m):
+            return loss_fn(m(x), y)
+        per_step_items = batch
+        unit = "images/sec"
+    net = net.to(device)
+    return net, step_fn, per_step_items, unit, batch
+
+
+def main():
+    args = parse_args()
+    if args.partition_mb:
+        os.environ["BPS_PARTITION_BYTES"] = str(args.partition_mb * 2**20)
+
+    import byteps_amd.torch as bps
+    bps.init()
+    world = bps.size()
+    rank = bps.rank()
+    device = torch.device(args.device, bps.local_rank()) \
+        if args.device == "cuda" else torch.device("cpu")
+    on_gpu = device.type == "cuda"
+    if on_gpu:
+        torch.backends.cudnn.benchmark = True
+
+    net, step_fn, per_step_items, unit, batch = build(args, device)
+
+    from byteps_amd.torch.parallel import DistributedDataParallel as DDP
+    model = DDP(net, broadcast_buffers=False)
+    opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
+                          weight_decay=1e-4)
+
+    use_bf16 = args.dtype == "bf16" and on_gpu
+    autocast = torch.autocast("cuda", dtype=torch.bfloat16) if use_bf16 \
+        else torch.autocast("cpu", enabled=False)
+
+    def train_step():
+        model.zero_grad_buckets()
+        with autocast:
+            loss = step_fn(model)
+        loss.backward()           # engine self-synchronizes on last grad
+        opt.step()
+        return loss
+
+    # -- hipGraph capture ---------------------------------------------------
+    graphed = None
+    want_graph = args.graph == "on" or (args.graph == "auto" and on_gpu)
+    if want_graph and on_gpu:
+        try:
+            for _ in range(3):      # warm up allocator + RCCL before capture
+                train_step()
+            torch.cuda.synchronize()
+            if world > 1:
+                dist.barrier()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                train_step()
+            graphed = g
+        except Exception as e:
+            if rank == 0:
+                print("graph capture failed (%s); falling back to eager"
+                      % type(e).__name__, file=sys.stderr)
+            graphed = None
+
+    def run_step():
+        if graphed is not None:
+            graphed.replay()
+        else:
+            train_step()
+
+    for _ in range(max(args.warmup, 1)):
+        run_step()
+
+    if world > 1:
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    if world > 1:
+        dist.barrier()
+
+    elapsed = t1 - t0
+    # MAX over ranks → aggregate uses the slowest rank's clock
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    value = per_step_items * world * args.steps / elapsed
+
+    if rank == 0:
+        out = {
+            "metric": unit.replace("/sec", "/sec (whole node)"),
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * world,
+                "seq_len": args.seq_len if args.model == "bert-large" else 224,
+                "parallelism": "dp%d" % world,
+                "graph": graphed is not None,
+                "compression": args.compression,
+            },
+        }
+        print(json.dumps(out))
+    bps.shutdown()
+
+
+if __name__ == "__main__":
+    main()

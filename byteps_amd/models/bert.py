@@ -1,0 +1,126 @@
+"""BERT (large) encoder for masked-LM pretraining benchmarks — the
+reference's headline scaling-efficiency number is BERT-large mixed
+precision (reference README.md:34-38, ≈334M params).
+
+MI355X-first implementation: attention goes through
+``F.scaled_dot_product_attention`` (ROCm flash/mem-efficient kernels),
+GELU fused via ``F.gelu(approximate='tanh')``, compute dtype bf16 under
+autocast.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class BertConfig:
+    def __init__(self, vocab_size=30522, hidden=1024, layers=24, heads=16,
+                 intermediate=4096, max_pos=512, dropout=0.0):
+        self.vocab_size = vocab_size
+        self.hidden = hidden
+        self.layers = layers
+        self.heads = heads
+        self.intermediate = intermediate
+        self.max_pos = max_pos
+        self.dropout = dropout
+
+    @staticmethod
+    def bert_large() -> "BertConfig":
+        return BertConfig(hidden=1024, layers=24, heads=16, intermediate=4096)
+
+    @staticmethod
+    def bert_base() -> "BertConfig":
+        return BertConfig(hidden=768, layers=12, heads=12, intermediate=3072)
+
+    @staticmethod
+    def tiny() -> "BertConfig":
+        return BertConfig(vocab_size=1024, hidden=64, layers=2, heads=4,
+                          intermediate=128, max_pos=128)
+
+
+class SelfAttention(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.heads = cfg.heads
+        self.head_dim = cfg.hidden // cfg.heads
+        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
+        self.out = nn.Linear(cfg.hidden, cfg.hidden)
+        self.dropout = cfg.dropout
+
+    def forward(self, x):
+        B, S, H = x.shape
+        qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
+        q, k, v = qkv.permute(2, 0, 3, 1, 4)          # 3 × (B, h, S, d)
+        o = F.scaled_dot_product_attention(
+            q, k, v, dropout_p=self.dropout if self.training else 0.0)
+        o = o.transpose(1, 2).reshape(B, S, H)
+        return self.out(o)
+
+
+class EncoderLayer(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.attn = SelfAttention(cfg)
+        self.ln1 = nn.LayerNorm(cfg.hidden)
+        self.fc1 = nn.Linear(cfg.hidden, cfg.intermediate)
+        self.fc2 = nn.Linear(cfg.intermediate, cfg.hidden)
+        self.ln2 = nn.LayerNorm(cfg.hidden)
+
+    def forward(self, x):
+        x = self.ln1(x + self.attn(x))
+        h = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
+        return self.ln2(x + h)
+
+
+class BertForPreTraining(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        self.pos_emb = nn.Embedding(cfg.max_pos, cfg.hidden)
+        self.type_emb = nn.Embedding(2, cfg.hidden)
+        self.emb_ln = nn.LayerNorm(cfg.hidden)
+        self.layers = nn.ModuleList(
+            EncoderLayer(cfg) for _ in range(cfg.layers))
+        self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
+        self.mlm_ln = nn.LayerNorm(cfg.hidden)
+        # decoder tied to token embedding
+        self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
+        self.apply(self._init)
+
+    def _init(self, m):
+        if isinstance(m, (nn.Linear, nn.Embedding)):
+            nn.init.normal_(m.weight, std=0.02)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.zeros_(m.bias)
+
+    def forward(self, input_ids, token_type_ids=None):
+        B, S = input_ids.shape
+        pos = torch.arange(S, device=input_ids.device).unsqueeze(0)
+        x = self.tok_emb(input_ids) + self.pos_emb(pos)
+        if token_type_ids is not None:
+            x = x + self.type_emb(token_type_ids)
+        x = self.emb_ln(x)
+        for layer in self.layers:
+            x = layer(x)
+        h = self.mlm_ln(F.gelu(self.mlm_dense(x), approximate="tanh"))
+        logits = F.linear(h, self.tok_emb.weight, self.mlm_bias)
+        return logits
+
+    def loss(self, input_ids, labels, token_type_ids=None):
+        logits = self(input_ids, token_type_ids)
+        return F.cross_entropy(
+            logits.view(-1, self.cfg.vocab_size).float(), labels.view(-1),
+            ignore_index=-100)
+
+
+def bert_large() -> BertForPreTraining:
+    return BertForPreTraining(BertConfig.bert_large())
+
+
+def bert_base() -> BertForPreTraining:
+    return BertForPreTraining(BertConfig.bert_base())

@@ -1,0 +1,309 @@
+// Gradient-compression codec kernels for gfx950 (CDNA4).
+//
+// The reference implemented every codec on the CPU (reference
+// common/compressor/impl/*.cc) and compressed *after* copying the full
+// gradient over PCIe.  Here codecs run on the GPU so only compressed
+// bytes cross PCIe/NIC; wavefront-wide ballots pack 64 sign bits per
+// instruction (64-wide waves — guide §1).
+//
+// Wire formats (defined by this framework, not reference-compatible):
+//   onebit:   uint64 words [ceil(n/64)] of sign bits (bit=1 ⇔ x>=0),
+//             + float scale_sum (Σ|x|; decompress divides by n)
+//   randomk:  k × (int32 index, float value), sampled counter-mode so the
+//             CPU server reproduces indices from (seed) alone
+//   dithering:int8 codes per element (linear: signed level in [-s, s];
+//             natural: signed exponent code, -128 ⇒ 0) + float norm
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include "common.h"
+
+#define BLOCK 256
+#define MAX_GRID 2048
+
+namespace {
+
+using bpsamd::rand_index;
+using bpsamd::uniform_at;
+
+inline int grid_for(int64_t work_items) {
+  int64_t blocks = (work_items + BLOCK - 1) / BLOCK;
+  return (int)(blocks < MAX_GRID ? (blocks > 0 ? blocks : 1) : MAX_GRID);
+}
+
+// ---------------------------------------------------------------------------
+// onebit (reference impl/onebit.cc:34-140): sign-pack + L1/n scale,
+// fused error-feedback update.
+// ---------------------------------------------------------------------------
+
+// Each wave packs 64 consecutive elements into one uint64 via __ballot.
+// Also accumulates the partial L1 sum (block-reduced, one atomic/block).
+__global__ void onebit_compress_kernel(const float* __restrict__ x, int64_t n,
+                                       unsigned long long* __restrict__ bits,
+                                       float* __restrict__ scale_sum) {
+  __shared__ float part[BLOCK / 64];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int waves_per_grid = (gridDim.x * BLOCK) >> 6;
+  const int64_t wave_id0 = (int64_t)blockIdx.x * (BLOCK >> 6) + wave;
+  const int64_t nwords = (n + 63) >> 6;
+  float l1 = 0.0f;
+  for (int64_t w = wave_id0; w < nwords; w += waves_per_grid) {
+    const int64_t i = (w << 6) + lane;
+    float v = (i < n) ? x[i] : -1.0f;   // pad lanes vote 0
+    l1 += (i < n) ? fabsf(v) : 0.0f;
+    unsigned long long mask = __ballot(v >= 0.0f);
+    if (lane == 0) bits[w] = mask;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) l1 += __shfl_down(l1, off, 64);
+  if (lane == 0) part[wave] = l1;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float b = part[0];
+    for (int w2 = 1; w2 < BLOCK / 64; ++w2) b += part[w2];
+    atomicAdd(scale_sum, b);
+  }
+}
+
+__global__ void onebit_decompress_kernel(
+    const unsigned long long* __restrict__ bits,
+    const float* __restrict__ scale_sum, int64_t n, float* __restrict__ out) {
+  const float scale = scale_sum[0] / (float)n;
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride) {
+    unsigned long long w = bits[i >> 6];
+    out[i] = ((w >> (i & 63)) & 1ULL) ? scale : -scale;
+  }
+}
+
+// error = input - decompress(compressed)  (reference fused update,
+// impl/onebit.cc:113-140)
+__global__ void onebit_error_kernel(const float* __restrict__ x,
+                                    const unsigned long long* __restrict__ bits,
+                                    const float* __restrict__ scale_sum,
+                                    int64_t n, float* __restrict__ err) {
+  const float scale = scale_sum[0] / (float)n;
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride) {
+    unsigned long long w = bits[i >> 6];
+    err[i] = x[i] - (((w >> (i & 63)) & 1ULL) ? scale : -scale);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// randomk (reference impl/randomk.cc:47-67): k pseudo-random (idx, val)
+// pairs.  Counter-mode RNG → thread j draws index j independently; the CPU
+// server regenerates the same indices from the seed.
+// ---------------------------------------------------------------------------
+
+__global__ void randomk_compress_kernel(const float* __restrict__ x, int64_t n,
+                                        int64_t k, uint64_t seed,
+                                        int32_t* __restrict__ idx,
+                                        float* __restrict__ val) {
+  int64_t j0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t j = j0; j < k; j += stride) {
+    int64_t i = (int64_t)rand_index(seed, (uint64_t)j, (uint64_t)n);
+    idx[j] = (int32_t)i;
+    val[j] = x[i];
+  }
+}
+
+// decompress: out (pre-zeroed) gets out[idx[j]] = val[j]; duplicate draws
+// write the same element's value — plain store is idempotent.
+__global__ void sparse_scatter_kernel(const int32_t* __restrict__ idx,
+                                      const float* __restrict__ val, int64_t k,
+                                      float* __restrict__ out) {
+  int64_t j0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t j = j0; j < k; j += stride) out[idx[j]] = val[j];
+}
+
+// error-feedback for sparse codecs: err = x, then err[idx[j]] = 0
+// (selected coordinates were transmitted exactly; reference topk/randomk
+// FastUpdateError semantics)
+__global__ void sparse_error_zero_kernel(const int32_t* __restrict__ idx,
+                                         int64_t k, float* __restrict__ err) {
+  int64_t j0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t j = j0; j < k; j += stride) err[idx[j]] = 0.0f;
+}
+
+// gather for topk packing (selection itself uses rocPRIM via torch.topk —
+// a library call, like GEMMs via hipBLASLt)
+__global__ void sparse_gather_kernel(const float* __restrict__ x,
+                                     const int32_t* __restrict__ idx,
+                                     int64_t k, float* __restrict__ val) {
+  int64_t j0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t j = j0; j < k; j += stride) val[j] = x[idx[j]];
+}
+
+// ---------------------------------------------------------------------------
+// dithering (reference impl/dithering.cc:51-121): stochastic quantization
+// against a norm, linear or natural (power-of-2) level partitions.
+// ---------------------------------------------------------------------------
+
+// linear: r = |x|/norm ∈ [0,1]; level = floor(r*s) + bernoulli(frac);
+// code = sign * level ∈ [-s, s] (s ≤ 127)
+__global__ void dithering_linear_compress_kernel(
+    const float* __restrict__ x, int64_t n, int s, uint64_t seed,
+    const float* __restrict__ norm, int8_t* __restrict__ code) {
+  const float nrm = norm[0];
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride) {
+    float v = x[i];
+    float r = (nrm > 0.0f) ? fabsf(v) / nrm : 0.0f;
+    float t = r * s;
+    int level = (int)t;
+    float frac = t - level;
+    level += (uniform_at(seed, (uint64_t)i) < frac) ? 1 : 0;
+    if (level > s) level = s;
+    code[i] = (int8_t)(v < 0.0f ? -level : level);
+  }
+}
+
+__global__ void dithering_linear_decompress_kernel(
+    const int8_t* __restrict__ code, int64_t n, int s,
+    const float* __restrict__ norm, float* __restrict__ out) {
+  const float nrm = norm[0];
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride)
+    out[i] = (float)code[i] / (float)s * nrm;
+}
+
+// natural: levels are powers of two.  r ∈ (0,1] lies in [2^e, 2^(e+1));
+// round stochastically to an endpoint, store signed biased exponent:
+// code = sign * (e + 128 + 1) clamped; 0 ⇒ value 0.  |e| ≤ 126.
+__global__ void dithering_natural_compress_kernel(
+    const float* __restrict__ x, int64_t n, uint64_t seed,
+    const float* __restrict__ norm, int8_t* __restrict__ code) {
+  const float nrm = norm[0];
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride) {
+    float v = x[i];
+    float r = (nrm > 0.0f) ? fabsf(v) / nrm : 0.0f;
+    if (r <= 0.0f) { code[i] = 0; continue; }
+    int e;
+    float m = frexpf(r, &e);         // r = m * 2^e, m ∈ [0.5, 1)
+    // interval endpoints: lo = 2^(e-1), hi = 2^e ; p(up) = (r-lo)/(lo)
+    float p_up = m * 2.0f - 1.0f;    // (r - lo)/lo
+    int ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
+    if (ebits < -120) { code[i] = 0; continue; }
+    if (ebits > 0) ebits = 0;        // r ≤ 1 ⇒ level ≤ 2^0
+    int biased = ebits + 121;        // ∈ [1, 121]
+    code[i] = (int8_t)(v < 0.0f ? -biased : biased);
+  }
+}
+
+__global__ void dithering_natural_decompress_kernel(
+    const int8_t* __restrict__ code, int64_t n, const float* __restrict__ norm,
+    float* __restrict__ out) {
+  const float nrm = norm[0];
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride) {
+    int c = code[i];
+    if (c == 0) { out[i] = 0.0f; continue; }
+    int mag = c < 0 ? -c : c;
+    float v = ldexpf(1.0f, mag - 121) * nrm;
+    out[i] = c < 0 ? -v : v;
+  }
+}
+
+}  // namespace
+
+#define STREAM reinterpret_cast<hipStream_t>(stream)
+
+extern "C" {
+
+int bps_onebit_compress(const void* x, int64_t n, void* bits, void* scale_sum,
+                        void* stream) {
+  int64_t nwords = (n + 63) >> 6;
+  hipLaunchKernelGGL(onebit_compress_kernel, dim3(grid_for(nwords * 64 / 4)),
+                     dim3(BLOCK), 0, STREAM, (const float*)x, n,
+                     (unsigned long long*)bits, (float*)scale_sum);
+  return (int)hipGetLastError();
+}
+
+int bps_onebit_decompress(const void* bits, const void* scale_sum, int64_t n,
+                          void* out, void* stream) {
+  hipLaunchKernelGGL(onebit_decompress_kernel, dim3(grid_for(n)), dim3(BLOCK),
+                     0, STREAM, (const unsigned long long*)bits,
+                     (const float*)scale_sum, n, (float*)out);
+  return (int)hipGetLastError();
+}
+
+int bps_onebit_error(const void* x, const void* bits, const void* scale_sum,
+                     int64_t n, void* err, void* stream) {
+  hipLaunchKernelGGL(onebit_error_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     STREAM, (const float*)x,
+                     (const unsigned long long*)bits, (const float*)scale_sum,
+                     n, (float*)err);
+  return (int)hipGetLastError();
+}
+
+int bps_randomk_compress(const void* x, int64_t n, int64_t k, uint64_t seed,
+                         void* idx, void* val, void* stream) {
+  hipLaunchKernelGGL(randomk_compress_kernel, dim3(grid_for(k)), dim3(BLOCK),
+                     0, STREAM, (const float*)x, n, k, seed, (int32_t*)idx,
+                     (float*)val);
+  return (int)hipGetLastError();
+}
+
+int bps_sparse_scatter(const void* idx, const void* val, int64_t k, void* out,
+                       void* stream) {
+  hipLaunchKernelGGL(sparse_scatter_kernel, dim3(grid_for(k)), dim3(BLOCK), 0,
+                     STREAM, (const int32_t*)idx, (const float*)val, k,
+                     (float*)out);
+  return (int)hipGetLastError();
+}
+
+int bps_sparse_error_zero(const void* idx, int64_t k, void* err, void* stream) {
+  hipLaunchKernelGGL(sparse_error_zero_kernel, dim3(grid_for(k)), dim3(BLOCK),
+                     0, STREAM, (const int32_t*)idx, k, (float*)err);
+  return (int)hipGetLastError();
+}
+
+int bps_sparse_gather(const void* x, const void* idx, int64_t k, void* val,
+                      void* stream) {
+  hipLaunchKernelGGL(sparse_gather_kernel, dim3(grid_for(k)), dim3(BLOCK), 0,
+                     STREAM, (const float*)x, (const int32_t*)idx, k,
+                     (float*)val);
+  return (int)hipGetLastError();
+}
+
+int bps_dithering_compress(const void* x, int64_t n, int s, uint64_t seed,
+                           int natural, const void* norm, void* code,
+                           void* stream) {
+  if (natural)
+    hipLaunchKernelGGL(dithering_natural_compress_kernel, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, STREAM, (const float*)x, n, seed,
+                       (const float*)norm, (int8_t*)code);
+  else
+    hipLaunchKernelGGL(dithering_linear_compress_kernel, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, STREAM, (const float*)x, n, s, seed,
+                       (const float*)norm, (int8_t*)code);
+  return (int)hipGetLastError();
+}
+
+int bps_dithering_decompress(const void* code, int64_t n, int s, int natural,
+                             const void* norm, void* out, void* stream) {
+  if (natural)
+    hipLaunchKernelGGL(dithering_natural_decompress_kernel, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, STREAM, (const int8_t*)code, n,
+                       (const float*)norm, (float*)out);
+  else
+    hipLaunchKernelGGL(dithering_linear_decompress_kernel, dim3(grid_for(n)),
+                       dim3(BLOCK), 0, STREAM, (const int8_t*)code, n, s,
+                       (const float*)norm, (float*)out);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -1,0 +1,228 @@
+// CPU reducer + CPU codec implementations (OpenMP).
+//
+// Used by the PS server process (which has no GPU — reference
+// common/cpu_reducer.cc:59-439, server/server.cc) and as the host-side
+// golden path for the HIP codecs: the RNG construction in common.h is
+// shared, so worker-GPU compression and server-CPU decompression agree
+// bit-for-bit on indices and stochastic rounding decisions.
+
+#include <algorithm>
+#include <cmath>
+#include <cstdint>
+#include <cstring>
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+#include "common.h"
+
+namespace bpsamd {
+
+static inline float bf16_to_f32(uint16_t v) {
+  uint32_t u = (uint32_t)v << 16;
+  float f;
+  std::memcpy(&f, &u, 4);
+  return f;
+}
+
+static inline uint16_t f32_to_bf16(float f) {
+  uint32_t u;
+  std::memcpy(&u, &f, 4);
+  // round-to-nearest-even
+  uint32_t rounding = 0x7FFF + ((u >> 16) & 1);
+  return (uint16_t)((u + rounding) >> 16);
+}
+
+extern "C" {
+
+// dst += src  (fp32 accumulate for bf16, like the reference's F16C fp16
+// path, common/cpu_reducer.cc:96-141)
+int bps_cpu_sum(void* dst, const void* src, int64_t n, int dtype) {
+  if (dtype == 0) {
+    float* d = (float*)dst;
+    const float* s = (const float*)src;
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i) d[i] += s[i];
+  } else if (dtype == 2) {
+    uint16_t* d = (uint16_t*)dst;
+    const uint16_t* s = (const uint16_t*)src;
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i)
+      d[i] = f32_to_bf16(bf16_to_f32(d[i]) + bf16_to_f32(s[i]));
+  } else if (dtype == 3) {
+    double* d = (double*)dst;
+    const double* s = (const double*)src;
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i) d[i] += s[i];
+  } else {
+    return -1;
+  }
+  return 0;
+}
+
+// dst = src1 + alpha * src2
+int bps_cpu_sum2(void* dst, const void* src1, const void* src2, int64_t n,
+                 float alpha, int dtype) {
+  if (dtype != 0) return -1;
+  float* d = (float*)dst;
+  const float* a = (const float*)src1;
+  const float* b = (const float*)src2;
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) d[i] = a[i] + alpha * b[i];
+  return 0;
+}
+
+int bps_cpu_copy(void* dst, const void* src, int64_t nbytes) {
+#pragma omp parallel
+  {
+    // split the copy across threads (reference OMP copy,
+    // common/cpu_reducer.cc:426-437)
+    int tid = 0, nthr = 1;
+#ifdef _OPENMP
+    tid = omp_get_thread_num();
+    nthr = omp_get_num_threads();
+#endif
+    int64_t chunk = (nbytes + nthr - 1) / nthr;
+    int64_t beg = tid * chunk;
+    int64_t end = std::min<int64_t>(nbytes, beg + chunk);
+    if (end > beg)
+      std::memcpy((char*)dst + beg, (const char*)src + beg, end - beg);
+  }
+  return 0;
+}
+
+int bps_cpu_scale(void* x, int64_t n, float alpha, int dtype) {
+  if (dtype == 0) {
+    float* d = (float*)x;
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i) d[i] *= alpha;
+    return 0;
+  }
+  return -1;
+}
+
+// -- CPU codecs (server side) ----------------------------------------------
+
+int bps_cpu_onebit_compress(const float* x, int64_t n, uint64_t* bits,
+                            float* scale_sum) {
+  int64_t nwords = (n + 63) >> 6;
+  double l1 = 0.0;
+#pragma omp parallel for reduction(+ : l1)
+  for (int64_t w = 0; w < nwords; ++w) {
+    uint64_t mask = 0;
+    int64_t lim = std::min<int64_t>(64, n - (w << 6));
+    for (int64_t l = 0; l < lim; ++l) {
+      float v = x[(w << 6) + l];
+      l1 += std::fabs(v);
+      if (v >= 0.0f) mask |= (1ULL << l);
+    }
+    bits[w] = mask;
+  }
+  *scale_sum = (float)l1;
+  return 0;
+}
+
+int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
+                              float* out) {
+  float scale = scale_sum / (float)n;
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i)
+    out[i] = ((bits[i >> 6] >> (i & 63)) & 1ULL) ? scale : -scale;
+  return 0;
+}
+
+int bps_cpu_randomk_indices(int64_t n, int64_t k, uint64_t seed,
+                            int32_t* idx) {
+  // regenerate the worker's draws from the seed (counter-mode)
+#pragma omp parallel for
+  for (int64_t j = 0; j < k; ++j)
+    idx[j] = (int32_t)rand_index(seed, (uint64_t)j, (uint64_t)n);
+  return 0;
+}
+
+int bps_cpu_sparse_scatter(const int32_t* idx, const float* val, int64_t k,
+                           float* out) {
+  for (int64_t j = 0; j < k; ++j) out[idx[j]] = val[j];
+  return 0;
+}
+
+// sparse sum into a dense accumulator: acc[idx[j]] += val[j]
+int bps_cpu_sparse_accumulate(const int32_t* idx, const float* val, int64_t k,
+                              float* acc) {
+  for (int64_t j = 0; j < k; ++j) acc[idx[j]] += val[j];
+  return 0;
+}
+
+int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
+                               int natural, float norm, int8_t* code) {
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) {
+    float v = x[i];
+    float r = (norm > 0.0f) ? std::fabs(v) / norm : 0.0f;
+    if (!natural) {
+      float t = r * s;
+      int level = (int)t;
+      float frac = t - level;
+      level += (uniform_at(seed, (uint64_t)i) < frac) ? 1 : 0;
+      if (level > s) level = s;
+      code[i] = (int8_t)(v < 0.0f ? -level : level);
+    } else {
+      if (r <= 0.0f) {
+        code[i] = 0;
+        continue;
+      }
+      int e;
+      float m = std::frexp(r, &e);
+      float p_up = m * 2.0f - 1.0f;
+      int ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
+      if (ebits < -120) {
+        code[i] = 0;
+        continue;
+      }
+      if (ebits > 0) ebits = 0;
+      int biased = ebits + 121;
+      code[i] = (int8_t)(v < 0.0f ? -biased : biased);
+    }
+  }
+  return 0;
+}
+
+int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
+                                 int natural, float norm, float* out) {
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) {
+    int c = code[i];
+    if (!natural) {
+      out[i] = (float)c / (float)s * norm;
+    } else {
+      if (c == 0) {
+        out[i] = 0.0f;
+        continue;
+      }
+      int mag = c < 0 ? -c : c;
+      float v = std::ldexp(1.0f, mag - 121) * norm;
+      out[i] = c < 0 ? -v : v;
+    }
+  }
+  return 0;
+}
+
+float bps_cpu_norm(const float* x, int64_t n, int mode) {
+  if (mode == 2) {
+    float mx = 0.0f;
+#pragma omp parallel for reduction(max : mx)
+    for (int64_t i = 0; i < n; ++i) mx = std::max(mx, std::fabs(x[i]));
+    return mx;
+  }
+  double acc = 0.0;
+#pragma omp parallel for reduction(+ : acc)
+  for (int64_t i = 0; i < n; ++i) {
+    float e = std::fabs(x[i]);
+    acc += (mode == 0) ? e : (double)e * e;
+  }
+  return (mode == 1) ? (float)std::sqrt(acc) : (float)acc;
+}
+
+}  // extern "C"
+}  // namespace bpsamd

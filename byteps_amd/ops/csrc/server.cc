@@ -1,0 +1,590 @@
+// byteps_amd parameter server (CPU) — from-scratch equivalent of the
+// reference's server process (reference server/server.cc:458-531,
+// server/queue.h), over the kv.h TCP protocol instead of ps-lite.
+//
+// Design (mirrors the reference's semantics, new implementation):
+//  - accept thread + one reader thread per worker connection;
+//  - N engine threads, each with a scheduling queue; keys are sharded to
+//    engine threads by accumulated load (reference server/server.h:154-178);
+//    with BPS_SERVER_ENABLE_SCHEDULE the queue pops the key with the
+//    fewest total pushes first — earliest layers first (reference
+//    server/queue.h:91-97);
+//  - per-key state machine: first push of a round COPIES into the fp32
+//    accumulator, later pushes SUM; when all expected pushers arrived the
+//    merge is versioned and queued pulls flush (reference
+//    server/server.cc:82-203,295-409);
+//  - compressed pushes are decompressed on the fly (CPU codecs shared
+//    with the HIP kernels via common.h RNG); the merged result is
+//    re-compressed once per round for the pull replies;
+//  - async mode sums straight into the store and answers pulls
+//    immediately (reference server/server.cc:315-319).
+
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <atomic>
+#include <condition_variable>
+#include <cstring>
+#include <deque>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <queue>
+#include <set>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+#include <pybind11/pybind11.h>
+
+#include "common.h"
+#include "kv.h"
+
+namespace py = pybind11;
+
+// CPU codec entry points (cpu_reducer.cc)
+extern "C" {
+int bps_cpu_sum(void* dst, const void* src, int64_t n, int dtype);
+int bps_cpu_onebit_compress(const float* x, int64_t n, uint64_t* bits,
+                            float* scale_sum);
+int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
+                              float* out);
+int bps_cpu_sparse_accumulate(const int32_t* idx, const float* val, int64_t k,
+                              float* acc);
+int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
+                               int natural, float norm, int8_t* code);
+int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
+                                 int natural, float norm, float* out);
+float bps_cpu_norm(const float* x, int64_t n, int mode);
+}
+
+namespace bpsamd {
+namespace {
+
+bool read_all_fd(int fd, void* buf, size_t n) {
+  char* p = (char*)buf;
+  while (n > 0) {
+    ssize_t r = ::read(fd, p, n);
+    if (r < 0) {
+      if (errno == EINTR) continue;
+      return false;
+    }
+    if (r == 0) return false;
+    p += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+struct Conn {
+  int fd;
+  std::mutex write_mu;
+
+  void send(const MsgHeader& h, const void* payload) {
+    std::lock_guard<std::mutex> lk(write_mu);
+    const char* pl = (const char*)payload;
+    size_t n = sizeof(MsgHeader);
+    const char* p = (const char*)&h;
+    while (n > 0) {
+      ssize_t w = ::write(fd, p, n);
+      if (w < 0) {
+        if (errno == EINTR) continue;
+        return;
+      }
+      p += w;
+      n -= (size_t)w;
+    }
+    n = h.len;
+    while (n > 0) {
+      ssize_t w = ::write(fd, pl, n);
+      if (w < 0) {
+        if (errno == EINTR) continue;
+        return;
+      }
+      pl += w;
+      n -= (size_t)w;
+    }
+  }
+};
+
+struct InitPayload {
+  uint64_t nelem;
+  uint32_t expected;
+  uint32_t levels;  // dithering s
+};
+
+struct PendingPull {
+  std::shared_ptr<Conn> conn;
+  MsgHeader hdr;
+};
+
+struct KeyState {
+  std::mutex mu;
+  std::vector<float> store;      // merged fp32 accumulator
+  std::vector<char> reply;       // compressed merged (codec mode)
+  uint64_t nelem = 0;
+  uint32_t expected = 1;
+  uint32_t codec = kRaw;
+  uint32_t levels = 64;
+  bool async_mode = false;
+  std::set<uint32_t> round_senders;
+  uint64_t version = 0;
+  std::vector<PendingPull> pending;
+  uint64_t push_total = 0;       // scheduling signal
+  std::vector<float> scratch;    // decompress workspace
+};
+
+struct Task {
+  std::shared_ptr<Conn> conn;
+  MsgHeader hdr;
+  std::vector<char> payload;
+  KeyState* ks;
+};
+
+class Server {
+ public:
+  Server(int port, int engine_threads, bool enable_schedule)
+      : engine_threads_(std::max(1, engine_threads)),
+        enable_schedule_(enable_schedule) {
+    listen_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
+    if (listen_fd_ < 0) throw std::runtime_error("server: socket() failed");
+    int one = 1;
+    setsockopt(listen_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in addr{};
+    addr.sin_family = AF_INET;
+    addr.sin_addr.s_addr = htonl(INADDR_ANY);
+    addr.sin_port = htons((uint16_t)port);
+    if (::bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0)
+      throw std::runtime_error("server: bind failed on port " +
+                               std::to_string(port));
+    socklen_t alen = sizeof(addr);
+    getsockname(listen_fd_, (sockaddr*)&addr, &alen);
+    port_ = ntohs(addr.sin_port);
+    if (::listen(listen_fd_, 64) != 0)
+      throw std::runtime_error("server: listen failed");
+  }
+
+  ~Server() { stop(); }
+
+  int port() const { return port_; }
+
+  void start() {
+    running_ = true;
+    for (int t = 0; t < engine_threads_; ++t) {
+      queues_.emplace_back(std::make_unique<EngineQueue>());
+      workers_.emplace_back([this, t] { engine_loop(t); });
+    }
+    accept_thread_ = std::thread([this] { accept_loop(); });
+  }
+
+  void stop() {
+    bool expected = true;
+    if (!running_.compare_exchange_strong(expected, false)) return;
+    ::shutdown(listen_fd_, SHUT_RDWR);
+    ::close(listen_fd_);
+    {
+      std::lock_guard<std::mutex> lk(conns_mu_);
+      for (auto& c : conns_) ::shutdown(c->fd, SHUT_RDWR);
+    }
+    if (accept_thread_.joinable()) accept_thread_.join();
+    for (auto& q : queues_) q->close();
+    for (auto& w : workers_) {
+      if (w.joinable()) w.join();
+    }
+    for (auto& r : readers_) {
+      if (r.joinable()) r.join();
+    }
+    {
+      std::lock_guard<std::mutex> lk(conns_mu_);
+      for (auto& c : conns_) ::close(c->fd);
+      conns_.clear();
+    }
+  }
+
+ private:
+  struct EngineQueue {
+    std::mutex mu;
+    std::condition_variable cv;
+    std::deque<Task> fifo;
+    bool closed = false;
+
+    void push(Task&& t) {
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        fifo.push_back(std::move(t));
+      }
+      cv.notify_one();
+    }
+    bool pop(Task& out, bool schedule) {
+      std::unique_lock<std::mutex> lk(mu);
+      cv.wait(lk, [this] { return closed || !fifo.empty(); });
+      if (fifo.empty()) return false;
+      if (!schedule) {
+        out = std::move(fifo.front());
+        fifo.pop_front();
+      } else {
+        // fewest-pushes-first (earliest layer) — reference server
+        // PriorityQueue (server/queue.h:91-97)
+        auto best = fifo.begin();
+        for (auto it = fifo.begin(); it != fifo.end(); ++it)
+          if (it->ks->push_total < best->ks->push_total) best = it;
+        out = std::move(*best);
+        fifo.erase(best);
+      }
+      return true;
+    }
+    void close() {
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        closed = true;
+      }
+      cv.notify_all();
+    }
+  };
+
+  void accept_loop() {
+    while (running_) {
+      int fd = ::accept(listen_fd_, nullptr, nullptr);
+      if (fd < 0) {
+        if (errno == EINTR) continue;
+        break;
+      }
+      int one = 1;
+      setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+      auto conn = std::make_shared<Conn>();
+      conn->fd = fd;
+      {
+        std::lock_guard<std::mutex> lk(conns_mu_);
+        conns_.push_back(conn);
+      }
+      readers_.emplace_back([this, conn] { reader_loop(conn); });
+    }
+  }
+
+  KeyState* key_state(uint64_t key) {
+    std::lock_guard<std::mutex> lk(keys_mu_);
+    auto it = keys_.find(key);
+    return it == keys_.end() ? nullptr : it->second.get();
+  }
+
+  void reader_loop(std::shared_ptr<Conn> conn) {
+    while (running_) {
+      MsgHeader h;
+      if (!read_all_fd(conn->fd, &h, sizeof(h))) return;
+      if (h.magic != kMagic) return;
+      std::vector<char> payload(h.len);
+      if (h.len > 0 && !read_all_fd(conn->fd, payload.data(), h.len)) return;
+
+      switch (h.op) {
+        case kInit:
+          handle_init(conn, h, payload);
+          break;
+        case kPush: {
+          KeyState* ks = key_state(h.key);
+          if (!ks) {
+            reply_err(conn, h);
+            break;
+          }
+          {
+            std::lock_guard<std::mutex> lk(ks->mu);
+            ks->push_total++;
+          }
+          int tid = engine_of(h.key);
+          queues_[tid]->push(Task{conn, h, std::move(payload), ks});
+          break;
+        }
+        case kPull: {
+          KeyState* ks = key_state(h.key);
+          if (!ks) {
+            reply_err(conn, h);
+            break;
+          }
+          handle_pull(conn, h, ks);
+          break;
+        }
+        case kBarrier:
+          handle_barrier(conn, h);
+          break;
+        case kShutdown:
+          return;
+        default:
+          return;
+      }
+    }
+  }
+
+  void reply_err(const std::shared_ptr<Conn>& conn, const MsgHeader& req) {
+    MsgHeader h = req;
+    h.len = 0;
+    h.aux = ~0ULL;  // error marker
+    h.op = req.op == kPull ? kPullReply : kPushReply;
+    conn->send(h, nullptr);
+  }
+
+  void handle_init(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
+                   const std::vector<char>& payload) {
+    if (payload.size() >= sizeof(InitPayload)) {
+      InitPayload ip;
+      std::memcpy(&ip, payload.data(), sizeof(ip));
+      std::lock_guard<std::mutex> lk(keys_mu_);
+      auto& slot = keys_[h.key];
+      if (!slot) {
+        slot = std::make_unique<KeyState>();
+        slot->nelem = ip.nelem;
+        slot->expected = std::max(1u, ip.expected);
+        slot->codec = cmd_codec(h.cmd);
+        slot->levels = std::max(1u, ip.levels);
+        slot->async_mode = cmd_async(h.cmd);
+        // first-init page-aligned store (reference
+        // server/server.cc:266-294); vector is 64 B aligned via resize
+        slot->store.assign(ip.nelem, 0.0f);
+        // engine assignment by least accumulated load (reference
+        // server/server.h:154-178)
+        int best = 0;
+        for (int t = 1; t < engine_threads_; ++t)
+          if (engine_load_[t] < engine_load_[best]) best = t;
+        engine_of_[h.key] = best;
+        engine_load_[best] += (int64_t)ip.nelem;
+      }
+    }
+    MsgHeader r = h;
+    r.op = kInitReply;
+    r.len = 0;
+    conn->send(r, nullptr);
+  }
+
+  int engine_of(uint64_t key) {
+    std::lock_guard<std::mutex> lk(keys_mu_);
+    auto it = engine_of_.find(key);
+    return it == engine_of_.end() ? 0 : it->second;
+  }
+
+  void handle_pull(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
+                   KeyState* ks) {
+    std::unique_lock<std::mutex> lk(ks->mu);
+    uint64_t want_version = h.aux;
+    if (ks->async_mode || ks->version >= want_version) {
+      send_pull_reply(conn, h, ks);
+    } else {
+      ks->pending.push_back(PendingPull{conn, h});
+    }
+  }
+
+  void send_pull_reply(const std::shared_ptr<Conn>& conn, const MsgHeader& req,
+                       KeyState* ks) {
+    // ks->mu held
+    MsgHeader r = req;
+    r.op = kPullReply;
+    r.aux = ks->version;
+    if (ks->codec == kRaw) {
+      r.len = ks->store.size() * sizeof(float);
+      conn->send(r, ks->store.data());
+    } else {
+      r.len = ks->reply.size();
+      conn->send(r, ks->reply.data());
+    }
+  }
+
+  void engine_loop(int tid) {
+    Task task;
+    while (queues_[tid]->pop(task, enable_schedule_)) {
+      process_push(task);
+    }
+  }
+
+  void process_push(Task& t) {
+    KeyState* ks = t.ks;
+    std::unique_lock<std::mutex> lk(ks->mu);
+    const uint32_t codec = cmd_codec(t.hdr.cmd);
+    const bool first = ks->async_mode ? false : ks->round_senders.empty();
+    const int64_t n = (int64_t)ks->nelem;
+    float* acc = ks->store.data();
+
+    switch (codec) {
+      case kRaw: {
+        const float* src = (const float*)t.payload.data();
+        if (first)
+          std::memcpy(acc, src, n * sizeof(float));
+        else
+          bps_cpu_sum(acc, src, n, 0);
+        break;
+      }
+      case kOnebit: {
+        int64_t nwords = (n + 63) >> 6;
+        float scale_sum;
+        std::memcpy(&scale_sum, t.payload.data() + nwords * 8, 4);
+        ks->scratch.resize(n);
+        bps_cpu_onebit_decompress((const uint64_t*)t.payload.data(),
+                                  scale_sum, n, ks->scratch.data());
+        if (first)
+          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
+        else
+          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        break;
+      }
+      case kTopk:
+      case kRandomk: {
+        int64_t k = (int64_t)t.hdr.aux & 0xFFFFFFFF;
+        const int32_t* idx = (const int32_t*)t.payload.data();
+        const float* val = (const float*)(t.payload.data() + k * 4);
+        if (first) std::memset(acc, 0, n * sizeof(float));
+        bps_cpu_sparse_accumulate(idx, val, k, acc);
+        break;
+      }
+      case kDitherLinear:
+      case kDitherNatural: {
+        float norm;
+        std::memcpy(&norm, t.payload.data(), 4);
+        const int8_t* code = (const int8_t*)(t.payload.data() + 4);
+        ks->scratch.resize(n);
+        bps_cpu_dithering_decompress(code, n, (int)ks->levels,
+                                     codec == kDitherNatural, norm,
+                                     ks->scratch.data());
+        if (first)
+          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
+        else
+          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        break;
+      }
+      default:
+        break;
+    }
+
+    // push ack
+    MsgHeader ack = t.hdr;
+    ack.op = kPushReply;
+    ack.len = 0;
+    t.conn->send(ack, nullptr);
+
+    if (ks->async_mode) return;
+
+    ks->round_senders.insert(t.hdr.sender);
+    if ((uint32_t)ks->round_senders.size() >= ks->expected) {
+      // ALL_RECV: finalize merge (reference server/server.cc:348-370)
+      ks->version++;
+      ks->round_senders.clear();
+      if (ks->codec != kRaw) compress_reply(ks);
+      auto pending = std::move(ks->pending);
+      ks->pending.clear();
+      for (auto& p : pending) {
+        if (p.hdr.aux <= ks->version) send_pull_reply(p.conn, p.hdr, ks);
+        else ks->pending.push_back(p);
+      }
+    }
+  }
+
+  void compress_reply(KeyState* ks) {
+    const int64_t n = (int64_t)ks->nelem;
+    const float* acc = ks->store.data();
+    switch (ks->codec) {
+      case kOnebit: {
+        int64_t nwords = (n + 63) >> 6;
+        ks->reply.resize(nwords * 8 + 8);
+        float scale_sum = 0.0f;
+        bps_cpu_onebit_compress(acc, n, (uint64_t*)ks->reply.data(),
+                                &scale_sum);
+        std::memcpy(ks->reply.data() + nwords * 8, &scale_sum, 4);
+        std::memset(ks->reply.data() + nwords * 8 + 4, 0, 4);
+        break;
+      }
+      case kTopk: {
+        int64_t k = std::max<int64_t>(1, std::min<int64_t>(n, (int64_t)ks->levels));
+        // k for topk is carried in `levels` at init
+        std::vector<int32_t> order(n);
+        for (int64_t i = 0; i < n; ++i) order[i] = (int32_t)i;
+        std::partial_sort(order.begin(), order.begin() + k, order.end(),
+                          [acc](int32_t a, int32_t b) {
+                            return std::fabs(acc[a]) > std::fabs(acc[b]);
+                          });
+        ks->reply.resize(k * 8);
+        int32_t* idx = (int32_t*)ks->reply.data();
+        float* val = (float*)(ks->reply.data() + k * 4);
+        for (int64_t j = 0; j < k; ++j) {
+          idx[j] = order[j];
+          val[j] = acc[order[j]];
+        }
+        break;
+      }
+      case kRandomk: {
+        int64_t k = std::max<int64_t>(1, std::min<int64_t>(n, (int64_t)ks->levels));
+        uint64_t seed = splitmix64(ks->version * 0x9E3779B97F4A7C15ULL + 11);
+        ks->reply.resize(k * 8);
+        int32_t* idx = (int32_t*)ks->reply.data();
+        float* val = (float*)(ks->reply.data() + k * 4);
+        for (int64_t j = 0; j < k; ++j) {
+          idx[j] = (int32_t)rand_index(seed, (uint64_t)j, (uint64_t)n);
+          val[j] = acc[idx[j]];
+        }
+        break;
+      }
+      case kDitherLinear:
+      case kDitherNatural: {
+        bool natural = ks->codec == kDitherNatural;
+        float norm = bps_cpu_norm(acc, n, natural ? 1 : 2);
+        uint64_t seed = splitmix64(ks->version * 0xD6E8FEB86659FD93ULL + 5);
+        ks->reply.resize(4 + n);
+        std::memcpy(ks->reply.data(), &norm, 4);
+        bps_cpu_dithering_compress(acc, n, (int)ks->levels, seed, natural,
+                                   norm, (int8_t*)(ks->reply.data() + 4));
+        break;
+      }
+      default:
+        break;
+    }
+  }
+
+  void handle_barrier(const std::shared_ptr<Conn>& conn, const MsgHeader& h) {
+    std::lock_guard<std::mutex> lk(barrier_mu_);
+    barrier_waiters_.push_back(PendingPull{conn, h});
+    uint32_t expected = (uint32_t)h.aux;
+    if (barrier_waiters_.size() >= expected) {
+      for (auto& w : barrier_waiters_) {
+        MsgHeader r = w.hdr;
+        r.op = kBarrierReply;
+        r.len = 0;
+        w.conn->send(r, nullptr);
+      }
+      barrier_waiters_.clear();
+    }
+  }
+
+  int listen_fd_;
+  int port_;
+  int engine_threads_;
+  bool enable_schedule_;
+  std::atomic<bool> running_{false};
+  std::thread accept_thread_;
+  std::vector<std::thread> readers_;
+  std::vector<std::thread> workers_;
+  std::vector<std::unique_ptr<EngineQueue>> queues_;
+  std::mutex conns_mu_;
+  std::vector<std::shared_ptr<Conn>> conns_;
+  std::mutex keys_mu_;
+  std::map<uint64_t, std::unique_ptr<KeyState>> keys_;
+  std::unordered_map<uint64_t, int> engine_of_;
+  std::map<int, int64_t> engine_load_;
+  std::mutex barrier_mu_;
+  std::vector<PendingPull> barrier_waiters_;
+};
+
+}  // namespace
+}  // namespace bpsamd
+
+void init_server(py::module_& m) {
+  using bpsamd::Server;
+  py::class_<Server>(m, "Server")
+      .def(py::init<int, int, bool>(), py::arg("port"),
+           py::arg("engine_threads") = 4, py::arg("enable_schedule") = false)
+      .def("start", &Server::start,
+           py::call_guard<py::gil_scoped_release>())
+      .def("stop", &Server::stop,
+           py::call_guard<py::gil_scoped_release>())
+      .def_property_readonly("port", &Server::port);
+}
