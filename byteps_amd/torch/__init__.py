@@ -56,7 +56,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
     def __init__(self, params, named_parameters=None,
                  compression=Compression.none,
                  backward_passes_per_step: int = 1,
-                 process_group=None):
+                 process_group=None,
+                 compression_params=None):
         super(self.__class__, self).__init__(params)
         self._compression = compression
         self.backward_passes_per_step = backward_passes_per_step
@@ -76,7 +77,8 @@ class _DistributedOptimizer(torch.optim.Optimizer):
                       else torch.float16 if compression is Compression.fp16
                       else None)
         self._engine = register_engine(GradEngine(
-            named, process_group=process_group, grad_dtype=grad_dtype))
+            named, process_group=process_group, grad_dtype=grad_dtype,
+            compression_params=compression_params))
         if backward_passes_per_step > 1:
             self._engine.set_sync_enabled(False)
 
@@ -109,16 +111,20 @@ class _DistributedOptimizer(torch.optim.Optimizer):
 def DistributedOptimizer(optimizer, named_parameters=None,
                          compression=Compression.none,
                          backward_passes_per_step: int = 1,
-                         process_group=None):
+                         process_group=None,
+                         compression_params=None):
     """Wrap ``optimizer`` for distributed training
-    (reference torch/__init__.py:226-265 dynamic subclassing pattern)."""
+    (reference torch/__init__.py:226-265 dynamic subclassing pattern).
+    ``compression_params`` selects a codec for the PS path, e.g.
+    ``{"compressor_type": "onebit", "ef_type": "vanilla"}`` (reference
+    mxnet compression_params, mxnet/__init__.py:250-317)."""
     cls = type(optimizer.__class__.__name__, (optimizer.__class__,),
                dict(_DistributedOptimizer.__dict__))
     obj = cls.__new__(cls)
     obj.__dict__.update(optimizer.__dict__)
     _DistributedOptimizer.__init__(
         obj, obj.param_groups, named_parameters, compression,
-        backward_passes_per_step, process_group)
+        backward_passes_per_step, process_group, compression_params)
     return obj
 
 

@@ -43,6 +43,21 @@ from ..common.partition import PartitionPlan, plan_partitions
 log = get_logger()
 
 
+def _is_dense_permutation(t: torch.Tensor) -> bool:
+    """True iff t's layout is a stride-permutation covering exactly
+    [0, numel) — e.g. channels_last — so a flat range can alias it."""
+    if t.numel() == 0:
+        return True
+    expect = 1
+    for s, sz in sorted(zip(t.stride(), t.shape)):
+        if sz == 1:
+            continue
+        if s != expect:
+            return False
+        expect *= sz
+    return True
+
+
 # --------------------------------------------------------------------------
 # Bucket: one schedulable flat buffer
 # --------------------------------------------------------------------------
@@ -95,7 +110,9 @@ class GradEngine:
         average: bool = True,
         grad_dtype: Optional[torch.dtype] = None,
         prescale: bool = False,
+        compression_params: Optional[dict] = None,
     ) -> None:
+        self.compression_params = dict(compression_params or {})
         C._require_init()
         self.cfg: Config = C.get_config()
         self.group = process_group
@@ -166,7 +183,15 @@ class GradEngine:
             bks = self.param_bucket[pidx]
             if len(bks) == 1 and self._single_span(bks[0], p) is not None:
                 view = self._single_span(bks[0], p)
-                p.grad = view.view_as(p)
+                # Alias with the PARAM's stride order (e.g. channels_last
+                # convs) so autograd accumulates without a layout
+                # conversion — the "gradient layout contract".  Any dense
+                # permuted layout is a bijection of the same flat range,
+                # so collectives on the flat buffer see the same bytes.
+                if not p.is_contiguous() and _is_dense_permutation(p):
+                    p.grad = view.as_strided(p.shape, p.stride())
+                else:
+                    p.grad = view.view_as(p)
             else:
                 # split across partitions: keep torch-allocated grad, copy
                 # into the views when the grad is produced

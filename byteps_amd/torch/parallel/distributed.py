@@ -28,7 +28,8 @@ class DistributedDataParallel(torch.nn.Module):
     def __init__(self, module: torch.nn.Module, device_ids=None,
                  broadcast_buffers: bool = True,
                  process_group=None,
-                 partition_bytes: Optional[int] = None):
+                 partition_bytes: Optional[int] = None,
+                 compression_params: Optional[dict] = None):
         super().__init__()
         _C._require_init()
         self.module = module
@@ -44,7 +45,8 @@ class DistributedDataParallel(torch.nn.Module):
         named = list(module.named_parameters())
         self._engine = register_engine(GradEngine(
             named, process_group=process_group,
-            partition_bytes=partition_bytes))
+            partition_bytes=partition_bytes,
+            compression_params=compression_params))
         self._engine.on_all_ready = self._self_synchronize
 
         # broadcast initial model state from rank 0 so replicas agree

@@ -1,19 +1,405 @@
-"""Hierarchical PS pipeline: reduce-scatter → push/pull KV → all-gather.
+"""Hierarchical PS pipeline — the inter-node push/pull path.
 
-Placeholder until the C++ KV transport lands (see DESIGN.md build order);
-PS mode is activated by ``BPS_NUM_SERVER>0`` or ``BPS_FORCE_DISTRIBUTED``.
+Route per bucket (reference queue-list construction,
+common/operations.cc:429-485, re-designed for MI355X):
+
+  intra-node RCCL reduce-scatter (xGMI)          [issued in backward-hook
+    → [compress shard — HIP codec kernel]         order, deterministic
+    → D2H of (compressed) shard into pinned       across ranks]
+      staging on a side HIP stream
+    → KV push → KV pull (C++ TCP client, GIL released)
+    → H2D → [decompress — HIP kernel]
+    → intra-node all-gather                      [issued at synchronize()
+                                                  in fixed bucket order]
+
+Only this rank's shard — compressed — crosses PCIe and the NIC; full
+gradients never leave HBM3E (the reference staged every byte through CPU
+shm even on one node, common/core_loops.cc:378-443).
+
+Collective-ordering rule: RCCL collectives on one communicator must be
+issued in the same order on every rank.  Reduce-scatters are issued from
+autograd hooks (same backward graph ⇒ same order — the same contract
+torch DDP relies on); all-gathers are issued in fixed bucket-index order
+during synchronize().  The middle (KV) section runs in a thread pool with
+per-task side streams and completes in any order.
 """
 
 from __future__ import annotations
 
+import struct
+import threading
+from concurrent.futures import Future, ThreadPoolExecutor
+from typing import Dict, List, Optional
 
-def get_pipeline(engine):
-    raise NotImplementedError(
-        "PS pipeline is not built yet — unset BPS_FORCE_DISTRIBUTED / "
-        "BPS_NUM_SERVER to use the RCCL-only path")
+import torch
+import torch.distributed as dist
+
+from .. import common as C
+from ..common import telemetry
+from ..common.logging_util import get_logger
+from ..common.naming import partition_key
+from ..compression import BaseCompressor, Compressed
+
+log = get_logger()
+
+_OP_INIT, _OP_PUSH, _OP_PULL = 5, 1, 2
 
 
-def get_tensor_pipeline():
-    raise NotImplementedError(
-        "PS pipeline is not built yet — unset BPS_FORCE_DISTRIBUTED / "
-        "BPS_NUM_SERVER to use the RCCL-only path")
+def _make_cmd(codec: int, dtype: int, async_mode: bool) -> int:
+    return (codec & 0xFF) | ((dtype & 0xFF) << 8) | (
+        (1 << 16) if async_mode else 0)
+
+
+class _KeyInfo:
+    __slots__ = ("key", "server", "nelem", "round", "initialized",
+                 "compressor")
+
+    def __init__(self, key: int, server: int, nelem: int,
+                 compressor: Optional[BaseCompressor]):
+        self.key = key
+        self.server = server
+        self.nelem = nelem
+        self.round = 0
+        self.initialized = False
+        self.compressor = compressor
+
+
+class _Staging:
+    """Pinned host buffers + side stream for one in-flight bucket."""
+
+    def __init__(self, nbytes_send: int, nbytes_recv: int, device):
+        self.on_gpu = device.type == "cuda"
+        pin = self.on_gpu
+        self.send = torch.empty(nbytes_send, dtype=torch.uint8,
+                                pin_memory=pin)
+        self.recv = torch.empty(nbytes_recv, dtype=torch.uint8,
+                                pin_memory=pin)
+        self.stream = torch.cuda.Stream(device) if self.on_gpu else None
+        self.gpu_scratch = None
+
+
+class Ticket:
+    __slots__ = ("future", "bucket", "shard", "done_event", "reply_view")
+
+    def __init__(self, future: Future, bucket, shard):
+        self.future = future
+        self.bucket = bucket
+        self.shard = shard
+        self.done_event = None
+        self.reply_view = None
+
+
+def _kv_client():
+    st = C._state
+    if st.kv is None:
+        from ..ops import core
+        from ..common.config import server_addresses
+        uris = server_addresses(st.cfg)
+        if not uris:
+            raise RuntimeError(
+                "PS mode needs BPS_SERVER_URIS=host:port[,host:port...] "
+                "(or start via the byteps_amd launcher)")
+        st.kv = core().KVClient(st.rank, uris)
+        if st.assigner is None:
+            from ..common.naming import ServerAssigner
+            st.assigner = ServerAssigner(len(uris))
+    return st.kv
+
+
+class PSPipeline:
+    """Per-GradEngine PS pipeline over the engine's buckets."""
+
+    def __init__(self, engine) -> None:
+        self.engine = engine
+        st = C._state
+        self.cfg = st.cfg
+        self.local_size = max(1, st.local_size)
+        self.world = engine.world
+        self.num_nodes = max(1, self.world // self.local_size)
+        self.node_id = st.rank // self.local_size
+        self.local_rank = st.rank % self.local_size
+        self.kv = _kv_client()
+        self.pool = ThreadPoolExecutor(
+            max_workers=max(2, self.cfg.compressor_threads),
+            thread_name_prefix="bps-ps")
+        self.keys: Dict[int, _KeyInfo] = {}
+        self._lock = threading.Lock()
+        self._staging: Dict[int, _Staging] = {}
+
+        # intra-node subgroup (reduce-scatter / all-gather scope).
+        self.node_group = None
+        if self.world > 1 and dist.is_initialized():
+            if self.num_nodes > 1:
+                for nid in range(self.num_nodes):
+                    ranks = list(range(nid * self.local_size,
+                                       (nid + 1) * self.local_size))
+                    g = dist.new_group(ranks)
+                    if nid == self.node_id:
+                        self.node_group = g
+            else:
+                self.node_group = None  # default group == node group
+        self.node_world = dist.get_world_size(self.node_group) \
+            if (self.world > 1 and dist.is_initialized()) else 1
+
+        # per-tensor compressor instances (engine-wide config for now;
+        # per-parameter params like the reference's byteps_* attrs can be
+        # layered on via set_compression_params)
+        self.compression_params: Dict = dict(
+            getattr(engine, "compression_params", {}) or {})
+
+    # -- helpers -----------------------------------------------------------
+
+    def _key_info(self, bucket) -> _KeyInfo:
+        with self._lock:
+            ki = self.keys.get(bucket.plan.index)
+            if ki is None:
+                from ..compression import create
+                nelem = bucket.buffer.numel() // self.node_world
+                pkey = partition_key(bucket.declared_key, self.local_rank)
+                nbytes = nelem * 4
+                server = C._state.assigner.assign(pkey, nbytes)
+                compressor = None
+                if self.compression_params.get("compressor_type") and \
+                        nbytes >= self.cfg.min_compress_bytes:
+                    compressor = create(dict(self.compression_params))
+                ki = _KeyInfo(pkey, server, nelem, compressor)
+                self.keys[bucket.plan.index] = ki
+            return ki
+
+    def _staging_for(self, bucket, ki: _KeyInfo) -> _Staging:
+        with self._lock:
+            s = self._staging.get(bucket.plan.index)
+            if s is None:
+                raw = ki.nelem * 4
+                if ki.compressor is not None:
+                    codec = ki.compressor.codec
+                    if codec in (2, 3):        # topk / randomk: 8 B per pair
+                        cap = 8 * min(max(ki.compressor.levels, 1), ki.nelem)
+                    elif codec == 1:           # onebit: bits + scale
+                        cap = ((ki.nelem + 63) // 64) * 8 + 8
+                    else:                      # dithering: norm + int8 codes
+                        cap = ki.nelem + 4
+                    cap = max(cap, 64)
+                else:
+                    cap = raw
+                s = _Staging(cap, cap, bucket.buffer.device)
+                self._staging[bucket.plan.index] = s
+            return s
+
+    def _ensure_init(self, ki: _KeyInfo) -> None:
+        if ki.initialized:
+            return
+        comp = ki.compressor
+        codec = comp.codec if comp is not None else 0
+        levels = comp.levels if comp is not None else 0
+        payload = struct.pack("<QII", ki.nelem, self.num_nodes, levels)
+        buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+        cmd = _make_cmd(codec, 0, self.cfg.enable_async)
+        t = self.kv.submit(ki.server, _OP_INIT, ki.key, buf.data_ptr(),
+                           len(payload), 0, 0, cmd, 0)
+        self.kv.wait(t)
+        ki.initialized = True
+
+    # -- main entry ---------------------------------------------------------
+
+    def submit(self, bucket) -> Ticket:
+        """Called from the autograd hook (deterministic order).  Issues the
+        intra-node reduce-scatter inline, then hands the KV section to the
+        pool."""
+        ki = self._key_info(bucket)
+        buf = bucket.buffer
+        dev = buf.device
+
+        if self.node_world > 1:
+            per = buf.numel() // self.node_world
+            shard = buf.narrow(0, self.local_rank * per, per)
+            if dist.get_backend(self.node_group) == "nccl":
+                dist.reduce_scatter_tensor(shard, buf, group=self.node_group)
+            else:
+                dist.all_reduce(buf, group=self.node_group)
+        else:
+            shard = buf
+
+        rs_event = None
+        if dev.type == "cuda":
+            rs_event = torch.cuda.Event()
+            rs_event.record(torch.cuda.current_stream(dev))
+
+        fut = self.pool.submit(self._kv_section, bucket, ki, shard, rs_event)
+        return Ticket(fut, bucket, shard)
+
+    def _kv_section(self, bucket, ki: _KeyInfo, shard: torch.Tensor,
+                    rs_event) -> tuple:
+        """Runs in a pool thread: compress → D2H → push/pull → H2D →
+        decompress.  Returns (done_event | None, reply_fp32 | None)."""
+        self._ensure_init(ki)
+        st = self._staging_for(bucket, ki)
+        comp = ki.compressor
+        ki.round += 1
+        on_gpu = shard.is_cuda
+
+        stream_ctx = torch.cuda.stream(st.stream) if on_gpu else _null_ctx()
+        with stream_ctx:
+            if on_gpu:
+                st.stream.wait_event(rs_event)
+            shard_f = shard if shard.dtype == torch.float32 \
+                else shard.float()
+            aux = ki.round
+            if comp is not None:
+                cp = comp.compress(shard_f)
+                payload = torch.cat(
+                    [p.reshape(-1).view(torch.uint8) for p in cp.parts])
+                nbytes = payload.numel()
+                push_aux = cp.aux if cp.aux else ki.round
+            else:
+                payload = shard_f.view(torch.uint8).reshape(-1)
+                nbytes = payload.numel()
+                push_aux = ki.round
+            st.send[:nbytes].copy_(payload, non_blocking=on_gpu)
+            if on_gpu:
+                st.stream.synchronize()
+
+        codec = comp.codec if comp is not None else 0
+        cmd = _make_cmd(codec, 0, self.cfg.enable_async)
+        t_push = self.kv.submit(ki.server, _OP_PUSH, ki.key,
+                                st.send.data_ptr(), nbytes, 0, 0, cmd,
+                                push_aux)
+        self.kv.wait(t_push)
+        telemetry.record(nbytes)
+        t_pull = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
+                                st.recv.data_ptr(), st.recv.numel(), cmd,
+                                ki.round)
+        reply_len, _ver = self.kv.wait(t_pull)
+        telemetry.record(reply_len)
+
+        done_event = None
+        with stream_ctx:
+            if comp is not None:
+                wire = st.recv[:reply_len]
+                if on_gpu:
+                    wire = wire.to(shard.device, non_blocking=True)
+                # sparse codecs: the server's reply k = min(levels, nelem);
+                # recover it from the wire length (8 B per pair)
+                reply_aux = reply_len // 8 if comp.codec in (2, 3) else 0
+                merged = comp.decompress(wire, ki.nelem, aux=reply_aux)
+                shard.copy_(merged.to(shard.dtype).reshape(shard.shape),
+                            non_blocking=on_gpu)
+            else:
+                host_f = st.recv[:reply_len].view(torch.float32)
+                if shard.dtype == torch.float32:
+                    shard.reshape(-1).copy_(host_f, non_blocking=on_gpu)
+                else:
+                    shard.reshape(-1).copy_(
+                        host_f.to(shard.dtype), non_blocking=on_gpu)
+            if on_gpu:
+                done_event = torch.cuda.Event()
+                done_event.record(st.stream)
+        return done_event, None
+
+    def wait(self, ticket: Ticket) -> None:
+        """Called from synchronize() in fixed bucket order: join the KV
+        future, chain the side stream into the main stream, and issue the
+        trailing all-gather (deterministic order)."""
+        done_event, _ = ticket.future.result()
+        buf = ticket.bucket.buffer
+        if done_event is not None:
+            torch.cuda.current_stream(buf.device).wait_event(done_event)
+        if self.node_world > 1:
+            if dist.get_backend(self.node_group) == "nccl":
+                dist.all_gather_into_tensor(buf, ticket.shard,
+                                            group=self.node_group)
+            else:
+                # chunks are contiguous views of buf — all_gather fills
+                # the bucket in place
+                chunks = list(buf.chunk(self.node_world))
+                dist.all_gather(chunks, ticket.shard.contiguous(),
+                                group=self.node_group)
+
+
+class _null_ctx:
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *a):
+        return False
+
+
+# -- standalone tensor pipeline (functional push_pull API in PS mode) -------
+
+class TensorWork:
+    def __init__(self, fut: Future):
+        self._fut = fut
+
+    def wait(self):
+        self._fut.result()
+
+    def is_completed(self):
+        return self._fut.done()
+
+
+class TensorPipeline:
+    """push_pull of an arbitrary tensor through the PS: every rank pushes
+    the whole tensor (expected pushers = world), the server sums, the pull
+    returns the global sum — the reference's non-bucketed push_pull path
+    (torch/ops.cc:54-97)."""
+
+    def __init__(self) -> None:
+        st = C._state
+        self.kv = _kv_client()
+        self.pool = ThreadPoolExecutor(max_workers=4,
+                                       thread_name_prefix="bps-pp")
+        self.world = st.size
+        self.rank = st.rank
+        self.keys: Dict[str, _KeyInfo] = {}
+        self._lock = threading.Lock()
+
+    def submit_tensor(self, tensor: torch.Tensor, name: str,
+                      priority: int = 0) -> TensorWork:
+        return TensorWork(self.pool.submit(self._run, tensor, name))
+
+    def _run(self, tensor: torch.Tensor, name: str):
+        st = C._state
+        key = st.registry.declare(name)
+        with self._lock:
+            ki = self.keys.get(name)
+            if ki is None:
+                pkey = partition_key(key, 0)
+                nelem = tensor.numel()
+                server = st.assigner.assign(pkey, nelem * 4)
+                ki = _KeyInfo(pkey, server, nelem, None)
+                self.keys[name] = ki
+        if not ki.initialized:
+            payload = struct.pack("<QII", ki.nelem, self.world, 0)
+            buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+            t = self.kv.submit(ki.server, _OP_INIT, ki.key, buf.data_ptr(),
+                               len(payload), 0, 0, 0, 0)
+            self.kv.wait(t)
+            ki.initialized = True
+        ki.round += 1
+        src = tensor.detach()
+        host = src.float().cpu().contiguous() if src.is_cuda \
+            else src.float().contiguous()
+        t = self.kv.submit(ki.server, _OP_PUSH, ki.key, host.data_ptr(),
+                           host.numel() * 4, 0, 0, 0, ki.round)
+        self.kv.wait(t)
+        recv = torch.empty_like(host)
+        t = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
+                           recv.data_ptr(), recv.numel() * 4, 0, ki.round)
+        self.kv.wait(t)
+        with torch.no_grad():
+            tensor.reshape(-1).copy_(
+                recv.to(tensor.device, tensor.dtype).reshape(-1))
+
+
+_tensor_pipeline: Optional[TensorPipeline] = None
+
+
+def get_pipeline(engine) -> PSPipeline:
+    return PSPipeline(engine)
+
+
+def get_tensor_pipeline() -> TensorPipeline:
+    global _tensor_pipeline
+    if _tensor_pipeline is None:
+        _tensor_pipeline = TensorPipeline()
+    return _tensor_pipeline

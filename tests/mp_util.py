@@ -42,7 +42,7 @@ def run_in_processes(fn, world: int, *args, extra_env=None, timeout: float = 120
     gloo-compatible rendezvous on 127.0.0.1.  Returns [result_rank0, ...].
     Raises on any rank failure."""
     ctx = mp.get_context("spawn")
-    q = ctx.SimpleQueue()
+    q = ctx.Queue()
     port = free_port()
     procs = []
     for r in range(world):
@@ -52,8 +52,13 @@ def run_in_processes(fn, world: int, *args, extra_env=None, timeout: float = 120
         procs.append(p)
     results = {}
     errors = []
+    import queue as _queue
     for _ in range(world):
-        rank, status, payload = q.get()
+        try:
+            rank, status, payload = q.get(timeout=timeout)
+        except _queue.Empty:
+            errors.append((-1, "harness timeout: a rank hung (deadlock?)"))
+            break
         if status == "ok":
             results[rank] = pickle.loads(payload)
         else:

@@ -1,0 +1,191 @@
+"""Hierarchical PS path tests: workers (gloo, CPU) + the native C++ KV
+server colocated on localhost — the reference's single-machine
+forced-distributed harness (tests/meta_test.py:27-85)."""
+
+import numpy as np
+import pytest
+import torch
+
+from mp_util import run_in_processes
+
+
+@pytest.fixture()
+def server():
+    from byteps_amd.ops import _core
+    srv = _core.Server(0, 2, False)
+    srv.start()
+    yield "127.0.0.1:%d" % srv.port
+    srv.stop()
+
+
+def _ps_env(uri, extra=None):
+    env = {"BPS_FORCE_DISTRIBUTED": "1", "BPS_SERVER_URIS": uri,
+           "BPS_NUM_SERVER": "1"}
+    env.update(extra or {})
+    return env
+
+
+def _make_model(seed=0):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 4))
+
+
+def _baseline(world, steps, lr=0.05):
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=lr)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = torch.cat(xs), torch.cat(ys)
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    return [p.detach().clone() for p in m.parameters()]
+
+
+def _ps_worker(rank, world, steps, partition_bytes):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    opt = bps.DistributedOptimizer(opt, named_parameters=m.named_parameters())
+    if partition_bytes:
+        for e in __import__("byteps_amd.torch.engine",
+                            fromlist=["_engines"])._engines:
+            pass
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = xs[rank], ys[rank]
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_ps_single_worker_parity(server):
+    """1 worker + 1 colocated server: full KV roundtrip, must equal local
+    SGD exactly (BASELINE.json config 1 plumbing)."""
+    expected = _baseline(1, 3)
+    results = run_in_processes(_ps_worker, 1, 3, None,
+                               extra_env=_ps_env(server))
+    for got, exp in zip(results[0], expected):
+        assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
+
+
+def test_ps_two_nodes_parity(server):
+    """2 simulated single-GPU nodes (local_size=1): server sums across
+    nodes; must equal big-batch baseline."""
+    expected = _baseline(2, 3)
+    results = run_in_processes(
+        _ps_worker, 2, 3, None,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1"}))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def test_ps_one_node_sharded_parity(server):
+    """2 ranks on ONE node: intra-node reduce + per-shard push/pull +
+    all-gather; must equal big-batch baseline."""
+    expected = _baseline(2, 3)
+    results = run_in_processes(_ps_worker, 2, 3, None,
+                               extra_env=_ps_env(server))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def _pp_tensor(rank, world):
+    import byteps_amd.torch as bps
+    bps.init()
+    t = torch.arange(100, dtype=torch.float32) + rank * 100
+    out = bps.push_pull(t, average=False, name="pp.ps")
+    expect = sum(torch.arange(100, dtype=torch.float32) + r * 100
+                 for r in range(world))
+    ok = torch.allclose(out, expect)
+    bps.shutdown()
+    return bool(ok)
+
+
+def test_ps_functional_push_pull(server):
+    assert all(run_in_processes(
+        _pp_tensor, 2, extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                                  "LOCAL_WORLD_SIZE": "1"})))
+
+
+def _ps_topk_full(rank, world, steps):
+    """topk with k == full size is lossless → exact parity through the
+    compressed wire format."""
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    opt = bps.DistributedOptimizer(
+        opt, named_parameters=m.named_parameters(),
+        compression_params={"compressor_type": "topk",
+                            "compressor_k": 1 << 20})
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = xs[rank], ys[rank]
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_ps_topk_lossless_parity(server):
+    expected = _baseline(2, 2)
+    results = run_in_processes(
+        _ps_topk_full, 2, 2,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1",
+                                   "BPS_MIN_COMPRESS_BYTES": "0"}))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-4, atol=1e-5)
+
+
+def _ps_onebit(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.01)
+    opt = bps.DistributedOptimizer(
+        opt, named_parameters=m.named_parameters(),
+        compression_params={"compressor_type": "onebit",
+                            "ef_type": "vanilla"})
+    torch.manual_seed(42)
+    x = torch.randn(16, 16)
+    y = torch.randn(16, 4)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = ((m(x) - y) ** 2).mean()
+        loss.backward()
+        losses.append(float(loss))
+        opt.step()
+    bps.shutdown()
+    return losses
+
+
+def test_ps_onebit_trains(server):
+    """onebit + EF through worker-GPU(CPU here)-server roundtrip: loss
+    must decrease (numerics are lossy by design)."""
+    results = run_in_processes(
+        _ps_onebit, 2, 30,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1",
+                                   "BPS_MIN_COMPRESS_BYTES": "0"}))
+    for losses in results:
+        assert losses[-1] < losses[0] * 0.9, losses[:3] + losses[-3:]
