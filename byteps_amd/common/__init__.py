@@ -137,9 +137,22 @@ def suspend() -> None:
     """Elastic training: release communication state but keep the declared
     tensor table so keys survive (reference byteps_suspend,
     common/operations.cc:96-107)."""
-    registry = _state.registry
-    shutdown()
-    _state.registry = registry
+    if not _state.initialized:
+        return
+    from ..torch import engine as torch_engine
+    torch_engine._suspend_engines()
+    if _state.kv is not None:
+        try:
+            _state.kv.close()
+        except Exception:
+            pass
+        _state.kv = None
+    if _state.tracer is not None:
+        _state.tracer.flush()
+    if _state.owns_process_group and dist.is_initialized():
+        dist.destroy_process_group()
+        _state.owns_process_group = False
+    _state.initialized = False
 
 
 def resume(num_workers: int, num_servers: int,
@@ -153,6 +166,8 @@ def resume(num_workers: int, num_servers: int,
         os.environ["RANK"] = str(global_rank)
     _state.registry.redeclare_all()
     init()
+    from ..torch import engine as torch_engine
+    torch_engine._resume_engines()
 
 
 def initialized() -> bool:

@@ -132,6 +132,10 @@ class GradEngine:
         # ready — DDP uses it to self-synchronize (reference
         # parallel/distributed.py:261-270)
         self.on_all_ready: Optional[Callable] = None
+        # invoked right after a bucket's collective is issued — the
+        # cross-barrier poller consumes this to apply per-bucket optimizer
+        # updates as results land (reference cross_barrier.py:159-186)
+        self.on_bucket_issued: Optional[Callable] = None
 
         named = [(n, p) for n, p in named_params if p.requires_grad]
         if not named:
@@ -273,17 +277,17 @@ class GradEngine:
         self._inflight_bytes += b.nbytes
         if C._state.tracer is not None:
             C._state.tracer.begin(b.declared_key, "comm", self._step)
-        if self.world <= 1 and self._ps is None:
-            b.work = None
-            return
-        if self.prescale:
-            b.buffer.div_(self.world)
-        if self._ps is not None:
-            b.ps_ticket = self._ps.submit(b)
-        elif self.world > 1:
-            b.work = dist.all_reduce(
-                b.buffer, op=dist.ReduceOp.SUM, group=self.group,
-                async_op=True)
+        if self.world > 1 or self._ps is not None:
+            if self.prescale:
+                b.buffer.div_(self.world)
+            if self._ps is not None:
+                b.ps_ticket = self._ps.submit(b)
+            else:
+                b.work = dist.all_reduce(
+                    b.buffer, op=dist.ReduceOp.SUM, group=self.group,
+                    async_op=True)
+        if self.on_bucket_issued is not None:
+            self.on_bucket_issued(b)
 
     # -- public API ---------------------------------------------------------
 
@@ -364,3 +368,20 @@ def _shutdown_engine() -> None:
         except Exception:
             pass
     _engines.clear()
+
+
+def _suspend_engines() -> None:
+    """Elastic suspend: detach hooks but keep engines registered so
+    resume() can re-arm them (reference byteps_suspend,
+    common/operations.cc:96-107)."""
+    for e in _engines:
+        e.detach()
+
+
+def _resume_engines() -> None:
+    for e in _engines:
+        e._attach_hooks()
+        e._ps = None
+        if C._state.ps_enabled:
+            from . import ps_pipeline
+            e._ps = ps_pipeline.get_pipeline(e)

@@ -1,0 +1,63 @@
+"""CrossBarrier: barrier-free pipelined updates must match plain
+synchronous SGD (reference cross_barrier.py semantics)."""
+
+import torch
+
+from mp_util import run_in_processes
+
+
+def _make_model(seed=0):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(
+        torch.nn.Linear(16, 32), torch.nn.ReLU(), torch.nn.Linear(32, 4))
+
+
+def _baseline(world, steps, lr=0.05, momentum=0.9):
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=lr, momentum=momentum)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = torch.cat(xs), torch.cat(ys)
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    return [p.detach().clone() for p in m.parameters()]
+
+
+def _cb_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    from byteps_amd.torch.cross_barrier import CrossBarrier
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9)
+    cb = CrossBarrier(m, opt)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = xs[rank], ys[rank]
+    for _ in range(steps):
+        cb.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        cb.step()
+    cb.synchronize()
+    out = [p.detach().clone() for p in m.parameters()]
+    cb.stop()
+    bps.shutdown()
+    return out
+
+
+def test_cross_barrier_single():
+    expected = _baseline(1, 5)
+    results = run_in_processes(_cb_worker, 1, 5)
+    for got, exp in zip(results[0], expected):
+        assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
+
+
+def test_cross_barrier_world2():
+    expected = _baseline(2, 5)
+    results = run_in_processes(_cb_worker, 2, 5)
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
