@@ -73,6 +73,16 @@ class _DistributedOptimizer(torch.optim.Optimizer):
         if len(named) != len({n for n, _ in named}):
             raise ValueError("named_parameters contains duplicate names")
 
+        self._async_ps = _C.get_config().enable_async and _C._state.ps_enabled
+        if self._async_ps:
+            # Asynchronous PS training (reference torch/__init__.py:195-223):
+            # no gradient sync — workers push weight deltas and pull back
+            # the server's running weights after every local step.
+            from .async_ps import AsyncPSWorker
+            self._engine = None
+            self._async_worker = AsyncPSWorker(named)
+            return
+
         grad_dtype = (torch.bfloat16 if compression is Compression.bf16
                       else torch.float16 if compression is Compression.fp16
                       else None)
@@ -83,14 +93,21 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._engine.set_sync_enabled(False)
 
     def zero_grad(self, set_to_none: bool = False):  # noqa: ARG002
+        if self._engine is None:
+            return super(self.__class__, self).zero_grad(set_to_none=False)
         # grads are views into persistent buckets — zero the buckets instead
         # of detaching (set_to_none would break the aliasing)
         self._engine.zero_grad()
 
     def synchronize(self):
-        self._engine.synchronize()
+        if self._engine is not None:
+            self._engine.synchronize()
 
     def step(self, closure=None):
+        if self._engine is None:      # async PS mode
+            loss = super(self.__class__, self).step(closure)
+            self._async_worker.exchange()
+            return loss
         self._pass_count += 1
         if self._pass_count < self.backward_passes_per_step:
             # accumulate only — no sync, no step

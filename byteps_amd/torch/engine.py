@@ -142,6 +142,7 @@ class GradEngine:
             raise ValueError("GradEngine: no parameters require grad")
         self.param_names = [n for n, _ in named]
         self.params = [p for _, p in named]
+        self._name_of = {id(p): n for n, p in named}
         for name in self.param_names:
             C._state.registry.declare("byteps.Gradient." + name)
 
@@ -277,6 +278,7 @@ class GradEngine:
         self._inflight_bytes += b.nbytes
         if C._state.tracer is not None:
             C._state.tracer.begin(b.declared_key, "comm", self._step)
+        self._debug_sample(b, "pre-comm")
         if self.world > 1 or self._ps is not None:
             if self.prescale:
                 b.buffer.div_(self.world)
@@ -288,6 +290,20 @@ class GradEngine:
                     async_op=True)
         if self.on_bucket_issued is not None:
             self.on_bucket_issued(b)
+
+    def _debug_sample(self, b: Bucket, stage: str) -> None:
+        """BPS_DEBUG_SAMPLE_TENSOR: print first/last values of a watched
+        tensor's bucket after each stage (reference
+        common/core_loops.cc:37-67)."""
+        watch = self.cfg.debug_sample_tensor
+        if not watch:
+            return
+        for p, g in zip(b.params, b.grads):
+            name = self._name_of.get(id(p), "?")
+            if watch in name:
+                log.info("[sample %s] %s: first=%g last=%g (bucket %d)",
+                         stage, name, float(g.reshape(-1)[0]),
+                         float(g.reshape(-1)[-1]), b.plan.index)
 
     # -- public API ---------------------------------------------------------
 
@@ -341,6 +357,9 @@ class GradEngine:
         if C._state.tracer is not None:
             for b in self.buckets:
                 C._state.tracer.end(b.declared_key, "comm", self._step)
+        if self.cfg.debug_sample_tensor:
+            for b in self.buckets:
+                self._debug_sample(b, "post-comm")
         for b in self.buckets:
             b.reset()
         self._ready_params = 0

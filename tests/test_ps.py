@@ -189,3 +189,39 @@ def test_ps_onebit_trains(server):
                                    "BPS_MIN_COMPRESS_BYTES": "0"}))
     for losses in results:
         assert losses[-1] < losses[0] * 0.9, losses[:3] + losses[-3:]
+
+
+def _async_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    torch.manual_seed(0)
+    m = torch.nn.Linear(16, 4)
+    opt = torch.optim.SGD(m.parameters(), lr=0.02)
+    opt = bps.DistributedOptimizer(opt, named_parameters=m.named_parameters())
+    torch.manual_seed(100 + rank)
+    x = torch.randn(32, 16)
+    w = torch.randn(16, 4)
+    y = x @ w
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = ((m(x) - y) ** 2).mean()
+        loss.backward()
+        losses.append(float(loss))
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return losses, out
+
+
+def test_ps_async_mode(server):
+    """BYTEPS_ENABLE_ASYNC: workers push weight deltas without a round
+    barrier; loss must decrease and replicas converge to the shared
+    server weights."""
+    results = run_in_processes(
+        _async_worker, 2, 40,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1",
+                                   "BPS_ENABLE_ASYNC": "1"}))
+    for losses, _params in results:
+        assert losses[-1] < losses[0] * 0.8, (losses[0], losses[-1])
