@@ -44,6 +44,26 @@ int bps_dithering_compress(const void* x, int64_t n, int s, uint64_t seed,
 int bps_dithering_decompress(const void* code, int64_t n, int s, int natural,
                              const void* norm, void* out, void* stream);
 
+// -- fused batchnorm (bn.hip) ----------------------------------------------
+int bps_bn_reduce(const void* x, long long M, int C, void* sums,
+                  void* stream);
+int bps_bn_finalize(const void* sums, long long M, int C, float eps,
+                    float momentum, void* mean_out, void* invstd_out,
+                    void* running_mean, void* running_var, int update_running,
+                    void* stream);
+int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
+                     int C, const void* mean, const void* invstd,
+                     const void* gamma, const void* beta, int relu,
+                     void* stream);
+int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
+                      long long M, int C, const void* mean,
+                      const void* invstd, void* sums2, int relu,
+                      void* stream);
+int bps_bn_bwd_apply(const void* x, const void* dy, const void* y, void* dx,
+                     void* dres, long long M, int C, const void* mean,
+                     const void* invstd, const void* gamma, const void* sums2,
+                     int relu, void* stream);
+
 // -- CPU reducer / codecs (cpu_reducer.cc) ---------------------------------
 int bps_cpu_sum(void* dst, const void* src, int64_t n, int dtype);
 int bps_cpu_sum2(void* dst, const void* src1, const void* src2, int64_t n,
@@ -156,6 +176,45 @@ PYBIND11_MODULE(_core, m) {
           check(bps_dithering_decompress(CP(code), n, sv, natural, CP(norm),
                                          P(out), P(s)),
                 "bps_dithering_decompress");
+        });
+
+  // fused batchnorm
+  m.def("bn_reduce", [](uintptr_t x, int64_t M, int C, uintptr_t sums,
+                        uintptr_t s) {
+    check(bps_bn_reduce(CP(x), M, C, P(sums), P(s)), "bps_bn_reduce");
+  });
+  m.def("bn_finalize",
+        [](uintptr_t sums, int64_t M, int C, float eps, float momentum,
+           uintptr_t mean, uintptr_t invstd, uintptr_t rmean, uintptr_t rvar,
+           int upd, uintptr_t s) {
+          check(bps_bn_finalize(CP(sums), M, C, eps, momentum, P(mean),
+                                P(invstd), P(rmean), P(rvar), upd, P(s)),
+                "bps_bn_finalize");
+        });
+  m.def("bn_fwd_apply",
+        [](uintptr_t x, uintptr_t res, uintptr_t y, int64_t M, int C,
+           uintptr_t mean, uintptr_t invstd, uintptr_t gamma, uintptr_t beta,
+           int relu, uintptr_t s) {
+          check(bps_bn_fwd_apply(CP(x), CP(res), P(y), M, C, CP(mean),
+                                 CP(invstd), CP(gamma), CP(beta), relu, P(s)),
+                "bps_bn_fwd_apply");
+        });
+  m.def("bn_bwd_reduce",
+        [](uintptr_t x, uintptr_t dy, uintptr_t y, int64_t M, int C,
+           uintptr_t mean, uintptr_t invstd, uintptr_t sums2, int relu,
+           uintptr_t s) {
+          check(bps_bn_bwd_reduce(CP(x), CP(dy), CP(y), M, C, CP(mean),
+                                  CP(invstd), P(sums2), relu, P(s)),
+                "bps_bn_bwd_reduce");
+        });
+  m.def("bn_bwd_apply",
+        [](uintptr_t x, uintptr_t dy, uintptr_t y, uintptr_t dx,
+           uintptr_t dres, int64_t M, int C, uintptr_t mean, uintptr_t invstd,
+           uintptr_t gamma, uintptr_t sums2, int relu, uintptr_t s) {
+          check(bps_bn_bwd_apply(CP(x), CP(dy), CP(y), P(dx), P(dres), M, C,
+                                 CP(mean), CP(invstd), CP(gamma), CP(sums2),
+                                 relu, P(s)),
+                "bps_bn_bwd_apply");
         });
 
   // CPU reducer / codecs
