@@ -32,8 +32,11 @@ def parse_args():
     p.add_argument("--model", default="resnet50",
                    choices=["resnet50", "vgg16", "bert-large", "mlp"])
     p.add_argument("--batch-size", type=int, default=0,
-                   help="per-GPU batch (default: 64 images / 8 seqs)")
-    p.add_argument("--seq-len", type=int, default=512)
+                   help="per-GPU batch (default 64 — the reference's "
+                        "published config, README.md:34-38)")
+    p.add_argument("--seq-len", type=int, default=128,
+                   help="BERT sequence length (128 = phase-1 pretraining, "
+                        "the regime of the reference's batch-64 runs)")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
                    help="hipGraph-capture the train step")
@@ -78,7 +81,7 @@ def build(args, device):
         unit = "images/sec"
     elif args.model == "bert-large":
         net = models.bert_large()
-        batch = args.batch_size or 8
+        batch = args.batch_size or 64
         S = args.seq_len
         ids = torch.randint(0, 30522, (batch, S), device=device)
         labels = torch.randint(0, 30522, (batch, S), device=device)

@@ -54,6 +54,32 @@ def djb2(key: int) -> int:
     return h
 
 
+def sdbm(key: int) -> int:
+    h = 0
+    for ch in str(key):
+        h = (ord(ch) + (h << 6) + (h << 16) - h) & 0xFFFFFFFFFFFFFFFF
+    return h
+
+
+def _naive(key: int) -> int:
+    return key
+
+
+_HASH_FNS = {"djb2": djb2, "sdbm": sdbm, "naive": _naive,
+             "built_in": lambda k: hash(str(k)) & 0xFFFFFFFFFFFFFFFF}
+
+
+def key_hash(key: int, fn: str = "") -> int:
+    """Pluggable key→server hash (reference BYTEPS_KEY_HASH_FN,
+    common/global.cc:628-677: naive/built-in/djb2/sdbm).  ``built_in``
+    is process-local (PYTHONHASHSEED-dependent) — use djb2/sdbm across
+    machines."""
+    import os
+    name = fn or os.environ.get(
+        "BPS_KEY_HASH_FN", os.environ.get("BYTEPS_KEY_HASH_FN", "djb2"))
+    return _HASH_FNS.get(name, djb2)(key)
+
+
 class NameRegistry:
     """Deterministic name → declared-key table.
 
@@ -117,7 +143,7 @@ class ServerAssigner:
     def assign(self, pkey: int, nbytes: int) -> int:
         if pkey in self.table:
             return self.table[pkey]
-        pref = djb2(pkey) % self.num_servers
+        pref = key_hash(pkey) % self.num_servers
         least = min(range(self.num_servers), key=lambda s: self.load[s])
         srv = pref if self.load[pref] - self.load[least] <= nbytes else least
         self.table[pkey] = srv

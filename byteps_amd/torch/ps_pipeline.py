@@ -119,6 +119,15 @@ class PSPipeline:
         self.node_id = st.rank // self.local_size
         self.local_rank = st.rank % self.local_size
         self.kv = _kv_client()
+        # alternative strategy (reference BYTEPS_REDUCE_ROOTS,
+        # common/global.cc:237-251): whole-bucket reduce to a key-hashed
+        # root rank + root-only push/pull + broadcast, instead of
+        # reduce-scatter sharding.  On xGMI the sharded path is usually
+        # faster (all 7 links busy); root-reduce helps when bucket count
+        # ≫ ranks and per-key latency dominates.
+        from ..common.config import env_bool
+        self.reduce_roots = env_bool("BPS_REDUCE_ROOTS",
+                                     "BYTEPS_REDUCE_ROOTS", default=False)
         self.pool = ThreadPoolExecutor(
             max_workers=max(2, self.cfg.compressor_threads),
             thread_name_prefix="bps-ps")
@@ -154,8 +163,12 @@ class PSPipeline:
             ki = self.keys.get(bucket.plan.index)
             if ki is None:
                 from ..compression import create
-                nelem = bucket.buffer.numel() // self.node_world
-                pkey = partition_key(bucket.declared_key, self.local_rank)
+                if self.reduce_roots:
+                    nelem = bucket.buffer.numel()
+                    pkey = partition_key(bucket.declared_key, 0)
+                else:
+                    nelem = bucket.buffer.numel() // self.node_world
+                    pkey = partition_key(bucket.declared_key, self.local_rank)
                 nbytes = nelem * 4
                 server = C._state.assigner.assign(pkey, nbytes)
                 compressor = None
@@ -210,7 +223,19 @@ class PSPipeline:
         buf = bucket.buffer
         dev = buf.device
 
-        if self.node_world > 1:
+        if self.node_world > 1 and self.reduce_roots:
+            root_local = bucket.declared_key % self.node_world
+            root_global = self.node_id * self.node_world + root_local
+            dist.reduce(buf, dst=root_global, group=self.node_group)
+            if self.local_rank != root_local:
+                # non-root: nothing to push; wait() broadcasts the result
+                fut = Future()
+                fut.set_result((None, None))
+                t = Ticket(fut, bucket, buf)
+                t.reply_view = ("bcast", root_global)
+                return t
+            shard = buf
+        elif self.node_world > 1:
             per = buf.numel() // self.node_world
             shard = buf.narrow(0, self.local_rank * per, per)
             if dist.get_backend(self.node_group) == "nccl":
@@ -237,6 +262,15 @@ class PSPipeline:
         comp = ki.compressor
         ki.round += 1
         on_gpu = shard.is_cuda
+        tracer = C._state.tracer
+        step = ki.round - 1
+
+        def _tr(stage, beginning):
+            if tracer is not None:
+                (tracer.begin if beginning else tracer.end)(
+                    ki.key, stage, step)
+
+        _tr("compress+d2h", True)
 
         stream_ctx = torch.cuda.stream(st.stream) if on_gpu else _null_ctx()
         with stream_ctx:
@@ -259,19 +293,25 @@ class PSPipeline:
             if on_gpu:
                 st.stream.synchronize()
 
+        _tr("compress+d2h", False)
         codec = comp.codec if comp is not None else 0
         cmd = _make_cmd(codec, 0, self.cfg.enable_async)
+        _tr("push", True)
         t_push = self.kv.submit(ki.server, _OP_PUSH, ki.key,
                                 st.send.data_ptr(), nbytes, 0, 0, cmd,
                                 push_aux)
         self.kv.wait(t_push)
+        _tr("push", False)
         telemetry.record(nbytes)
+        _tr("pull", True)
         t_pull = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
                                 st.recv.data_ptr(), st.recv.numel(), cmd,
                                 ki.round)
         reply_len, _ver = self.kv.wait(t_pull)
+        _tr("pull", False)
         telemetry.record(reply_len)
 
+        _tr("h2d+decompress", True)
         done_event = None
         with stream_ctx:
             if comp is not None:
@@ -294,6 +334,7 @@ class PSPipeline:
             if on_gpu:
                 done_event = torch.cuda.Event()
                 done_event.record(st.stream)
+        _tr("h2d+decompress", False)
         return done_event, None
 
     def wait(self, ticket: Ticket) -> None:
@@ -304,6 +345,11 @@ class PSPipeline:
         buf = ticket.bucket.buffer
         if done_event is not None:
             torch.cuda.current_stream(buf.device).wait_event(done_event)
+        if self.reduce_roots and self.node_world > 1:
+            root_local = ticket.bucket.declared_key % self.node_world
+            root_global = self.node_id * self.node_world + root_local
+            dist.broadcast(buf, src=root_global, group=self.node_group)
+            return
         if self.node_world > 1:
             if dist.get_backend(self.node_group) == "nccl":
                 dist.all_gather_into_tensor(buf, ticket.shard,

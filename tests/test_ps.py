@@ -225,3 +225,15 @@ def test_ps_async_mode(server):
                                    "BPS_ENABLE_ASYNC": "1"}))
     for losses, _params in results:
         assert losses[-1] < losses[0] * 0.8, (losses[0], losses[-1])
+
+
+def test_ps_reduce_roots_parity(server):
+    """BPS_REDUCE_ROOTS alternative strategy: whole-bucket reduce to a
+    hashed root + root-only push/pull + broadcast."""
+    expected = _baseline(2, 3)
+    results = run_in_processes(
+        _ps_worker, 2, 3, None,
+        extra_env=_ps_env(server, {"BPS_REDUCE_ROOTS": "1"}))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)

@@ -71,12 +71,15 @@ def test_ps_gpu_worker(mode, tmp_path):
     try:
         script = tmp_path / "worker.py"
         script.write_text(_WORKER)
+        repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
         env = dict(os.environ)
         env.update({
             "BPS_FORCE_DISTRIBUTED": "1",
             "BPS_NUM_SERVER": "1",
             "BPS_SERVER_URIS": "127.0.0.1:%d" % srv.port,
             "BPS_MIN_COMPRESS_BYTES": "0",
+            "PYTHONPATH": repo_root + os.pathsep + os.environ.get(
+                "PYTHONPATH", ""),
         })
         out = subprocess.run(
             [sys.executable, str(script), mode], env=env,
