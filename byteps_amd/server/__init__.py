@@ -6,6 +6,7 @@ from __future__ import annotations
 
 import os
 import signal
+import socket
 import threading
 
 from ..common.config import Config
@@ -26,6 +27,18 @@ def run_server(port: int = 0, block: bool = True):
     log.info("byteps_amd server listening on :%d (%d engine threads, "
              "schedule=%s)", srv.port, cfg.server_engine_threads,
              cfg.server_enable_schedule)
+    # announce to the scheduler if one is configured (reference ps-lite
+    # scheduler rendezvous; static BPS_SERVER_URIS needs no announcement)
+    if os.environ.get("BPS_ROOT_URI", os.environ.get("DMLC_PS_ROOT_URI")):
+        try:
+            from ..launcher.scheduler import register_server
+            host = os.environ.get("BPS_SERVER_HOST") or \
+                socket.gethostbyname(socket.gethostname())
+            register_server(host, srv.port, cfg.root_uri, cfg.root_port)
+            log.info("registered %s:%d with scheduler %s:%d", host,
+                     srv.port, cfg.root_uri, cfg.root_port)
+        except Exception as e:
+            log.warning("scheduler registration failed: %s", e)
     if block:
         stop = threading.Event()
         signal.signal(signal.SIGTERM, lambda *a: stop.set())

@@ -95,10 +95,15 @@ def _kv_client():
         from ..ops import core
         from ..common.config import server_addresses
         uris = server_addresses(st.cfg)
+        if not uris and st.cfg.num_servers > 0:
+            # scheduler rendezvous (reference DMLC_PS_ROOT_URI path)
+            from ..launcher.scheduler import discover_servers
+            uris = discover_servers(st.cfg.root_uri, st.cfg.root_port)
         if not uris:
             raise RuntimeError(
-                "PS mode needs BPS_SERVER_URIS=host:port[,host:port...] "
-                "(or start via the byteps_amd launcher)")
+                "PS mode needs BPS_SERVER_URIS=host:port[,...] or a "
+                "scheduler at BPS_ROOT_URI:BPS_ROOT_PORT with "
+                "BPS_NUM_SERVER set")
         st.kv = core().KVClient(st.rank, uris)
         if st.assigner is None:
             from ..common.naming import ServerAssigner
