@@ -159,6 +159,14 @@ def main():
                 print("graph capture failed (%s); falling back to eager"
                       % type(e).__name__, file=sys.stderr)
             graphed = None
+        if world > 1:
+            # replaying a captured collective sequence on only SOME ranks
+            # deadlocks — use the graph only if every rank captured
+            ok = torch.tensor([1.0 if graphed is not None else 0.0],
+                              device=device)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 0.5:
+                graphed = None
 
     def run_step():
         if graphed is not None:
