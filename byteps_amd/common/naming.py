@@ -14,7 +14,7 @@ Reproduces the reference's naming / key-space layout:
 
 from __future__ import annotations
 
-from typing import Dict, List
+from typing import Dict, List, Optional
 
 PART_BITS = 16
 MAX_PARTS = 1 << PART_BITS
@@ -126,26 +126,51 @@ class NameRegistry:
 
 
 class ServerAssigner:
-    """Key → server with djb2 hash + greedy load balancing.
+    """Key → server with hash + greedy weighted load balancing.
 
-    The reference hashes first and falls back to accumulated-load accounting
-    (common/global.cc:660-667).  We hash to a preferred server and accept it
-    unless its accumulated bytes exceed the least-loaded server by more than
-    one partition, in which case the least-loaded server takes the key —
-    deterministic given identical declaration sequences on all workers.
+    The reference hashes first and falls back to accumulated-load
+    accounting (common/global.cc:660-667); mixed mode gives colocated
+    servers (those sharing a machine with a worker) a smaller share
+    (common/global.cc:566-596).  Here: each server has a capacity weight;
+    a key goes to its hash-preferred server unless that server's
+    *effective* load (bytes/weight) exceeds the least-loaded one by more
+    than the partition size — deterministic given identical declaration
+    sequences on all workers.
+
+    Mixed mode: ``BPS_ENABLE_MIXED_MODE=1`` +
+    ``BPS_COLOCATED_SERVERS=i,j,...`` (indices into the server list);
+    colocated servers get weight ``BPS_MIXED_RATIO`` (default 0.5).
     """
 
-    def __init__(self, num_servers: int) -> None:
+    def __init__(self, num_servers: int,
+                 weights: Optional[List[float]] = None) -> None:
+        import os
         self.num_servers = max(1, num_servers)
-        self.load = [0] * self.num_servers
+        if weights is None:
+            weights = [1.0] * self.num_servers
+            if os.environ.get("BPS_ENABLE_MIXED_MODE",
+                              os.environ.get("BYTEPS_ENABLE_MIXED_MODE",
+                                             "0")) == "1":
+                ratio = float(os.environ.get("BPS_MIXED_RATIO", "0.5"))
+                colocated = os.environ.get("BPS_COLOCATED_SERVERS", "")
+                for tok in colocated.split(","):
+                    tok = tok.strip()
+                    if tok.isdigit() and int(tok) < self.num_servers:
+                        weights[int(tok)] = ratio
+        self.weights = weights
+        self.load = [0.0] * self.num_servers
         self.table: Dict[int, int] = {}
+
+    def _eff(self, s: int) -> float:
+        return self.load[s] / max(self.weights[s], 1e-9)
 
     def assign(self, pkey: int, nbytes: int) -> int:
         if pkey in self.table:
             return self.table[pkey]
         pref = key_hash(pkey) % self.num_servers
-        least = min(range(self.num_servers), key=lambda s: self.load[s])
-        srv = pref if self.load[pref] - self.load[least] <= nbytes else least
+        least = min(range(self.num_servers), key=self._eff)
+        srv = pref if self._eff(pref) - self._eff(least) <= \
+            nbytes / max(self.weights[pref], 1e-9) else least
         self.table[pkey] = srv
         self.load[srv] += nbytes
         return srv

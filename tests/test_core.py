@@ -92,3 +92,15 @@ def test_config_partition_rounding(monkeypatch):
     monkeypatch.setenv("BPS_LOCAL_SIZE", "8")
     c = cfg_mod.Config.from_env()
     assert c.partition_bytes % (8 * cfg_mod.PAGE_SIZE) == 0
+
+
+def test_mixed_mode_weighted_assignment(monkeypatch):
+    monkeypatch.setenv("BPS_ENABLE_MIXED_MODE", "1")
+    monkeypatch.setenv("BPS_COLOCATED_SERVERS", "0")
+    monkeypatch.setenv("BPS_MIXED_RATIO", "0.5")
+    a = ServerAssigner(2)
+    for k in range(400):
+        a.assign(k, 1000)
+    # colocated server 0 gets ~half the standalone server's bytes
+    ratio = a.load[0] / a.load[1]
+    assert 0.4 < ratio < 0.65, ratio
