@@ -49,15 +49,22 @@ __device__ inline void store8(bf16* p, const float* in) {
   *reinterpret_cast<short8*>(p) = r;
 }
 
+// Fixed partial-buffer depth: every reduce launches exactly RED_BLOCKS
+// blocks (1024 waves — enough to saturate HBM) writing per-block partial
+// sums; the finalize kernel folds the partials.  Per-address atomicAdd
+// chains from a 2048-block grid measured 431 µs/call (profile round 2) —
+// partials + a folding pass run at memory speed.
+#define RED_BLOCKS 256
+
 // ---------------------------------------------------------------------------
-// fwd reduce: sums[c] = Σ_m x[m,c] ; sums[C+c] = Σ_m x[m,c]²
+// fwd reduce: partial[b][c] = Σ_rows(b) x[m,c] ; partial[b][C+c] = Σ x²
 // Each thread owns 8 consecutive channels (one short8 column slice) and a
-// row subset; LDS-reduce across the row groups, one atomicAdd per 8
-// channels per block.
+// row subset; LDS-reduce across the row groups, leader writes the
+// block's partial (no atomics).
 // ---------------------------------------------------------------------------
 
 __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
-                                 int C, float* __restrict__ sums) {
+                                 int C, float* __restrict__ partial) {
   const int cpt = C >> 3;                        // short8 slots per row
   const int groups = max(1, BLOCK / cpt);        // rows handled per pass
   const int t = threadIdx.x;
@@ -81,6 +88,7 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
     }
   }
   // LDS reduce across groups for this c8 (sum pass, then sq pass)
+  float* out = partial + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int i = 0; i < 8; ++i) lds[t * 8 + i] = s.v[i];
   __syncthreads();
@@ -89,7 +97,7 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) atomicAdd(&sums[(c8 << 3) + i], s.v[i]);
+    for (int i = 0; i < 8; ++i) out[(c8 << 3) + i] = s.v[i];
   }
   __syncthreads();
 #pragma unroll
@@ -100,8 +108,7 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
 #pragma unroll
       for (int i = 0; i < 8; ++i) q.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
-      atomicAdd(&sums[C + (c8 << 3) + i], q.v[i]);
+    for (int i = 0; i < 8; ++i) out[C + (c8 << 3) + i] = q.v[i];
   }
 }
 
@@ -109,7 +116,7 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
 // finalize: mean/invstd from sums; running-stat update (1 thread/channel)
 // ---------------------------------------------------------------------------
 
-__global__ void bn_finalize_kernel(const float* __restrict__ sums,
+__global__ void bn_finalize_kernel(const float* __restrict__ partial,
                                    long long M, int C, float eps,
                                    float momentum,
                                    float* __restrict__ mean_out,
@@ -119,9 +126,14 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sums,
                                    int update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float sum = 0.0f, sumsq = 0.0f;
+  for (int b = 0; b < RED_BLOCKS; ++b) {
+    sum += partial[(long long)b * 2 * C + c];
+    sumsq += partial[(long long)b * 2 * C + C + c];
+  }
   float n = (float)M;
-  float mean = sums[c] / n;
-  float var = sums[C + c] / n - mean * mean;
+  float mean = sum / n;
+  float var = sumsq / n - mean * mean;
   if (var < 0.0f) var = 0.0f;
   mean_out[c] = mean;
   invstd_out[c] = rsqrtf(var + eps);
@@ -169,7 +181,8 @@ __global__ void bn_fwd_apply_kernel(const bf16* __restrict__ x,
 
 // ---------------------------------------------------------------------------
 // bwd reduce: dz = dy masked by y>0 (if RELU);
-//   sums2[c] = Σ dz ; sums2[C+c] = Σ dz * xhat
+//   partial[b][c] = Σ_rows(b) dz ; partial[b][C+c] = Σ dz * xhat
+// folded by bn_fold_kernel into sums2[2C].
 // ---------------------------------------------------------------------------
 
 template <bool RELU>
@@ -178,7 +191,7 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
                                      const bf16* __restrict__ y, long long M,
                                      int C, const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
-                                     float* __restrict__ sums2) {
+                                     float* __restrict__ partial) {
   const int cpt = C >> 3;
   const int groups = max(1, BLOCK / cpt);
   const int t = threadIdx.x;
@@ -207,6 +220,7 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
       }
     }
   }
+  float* out = partial + (long long)blockIdx.x * 2 * C;
 #pragma unroll
   for (int i = 0; i < 8; ++i) lds[t * 8 + i] = s1.v[i];
   __syncthreads();
@@ -215,7 +229,7 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s1.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) atomicAdd(&sums2[(c8 << 3) + i], s1.v[i]);
+    for (int i = 0; i < 8; ++i) out[(c8 << 3) + i] = s1.v[i];
   }
   __syncthreads();
 #pragma unroll
@@ -226,9 +240,19 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s2.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
-      atomicAdd(&sums2[C + (c8 << 3) + i], s2.v[i]);
+    for (int i = 0; i < 8; ++i) out[C + (c8 << 3) + i] = s2.v[i];
   }
+}
+
+// fold the RED_BLOCKS partials into sums2[2C]
+__global__ void bn_fold_kernel(const float* __restrict__ partial, int C,
+                               float* __restrict__ sums2) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= 2 * C) return;
+  float acc = 0.0f;
+  for (int b = 0; b < RED_BLOCKS; ++b)
+    acc += partial[(long long)b * 2 * C + c];
+  sums2[c] = acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -289,12 +313,13 @@ inline int grid_for_elems(long long elems) {
 
 extern "C" {
 
-int bps_bn_reduce(const void* x, long long M, int C, void* sums,
+int bps_bn_red_blocks(void) { return RED_BLOCKS; }
+
+int bps_bn_reduce(const void* x, long long M, int C, void* partial,
                   void* stream) {
   if ((C & 7) || C > 2048) return -1;
-  hipLaunchKernelGGL(bn_reduce_kernel, dim3(grid_for_rows(M, C)),
-                     dim3(BLOCK), 0, STREAM, (const bf16*)x, M, C,
-                     (float*)sums);
+  hipLaunchKernelGGL(bn_reduce_kernel, dim3(RED_BLOCKS), dim3(BLOCK), 0,
+                     STREAM, (const bf16*)x, M, C, (float*)partial);
   return (int)hipGetLastError();
 }
 
@@ -332,20 +357,28 @@ int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
 
 int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
                       long long M, int C, const void* mean,
-                      const void* invstd, void* sums2, int relu,
+                      const void* invstd, void* partial, int relu,
                       void* stream) {
   if ((C & 7) || C > 2048) return -1;
-  int g = grid_for_rows(M, C);
   if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(g), dim3(BLOCK), 0,
-                       STREAM, (const bf16*)x, (const bf16*)dy,
-                       (const bf16*)y, M, C, (const float*)mean,
-                       (const float*)invstd, (float*)sums2);
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(RED_BLOCKS),
+                       dim3(BLOCK), 0, STREAM, (const bf16*)x,
+                       (const bf16*)dy, (const bf16*)y, M, C,
+                       (const float*)mean, (const float*)invstd,
+                       (float*)partial);
   else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(g), dim3(BLOCK),
-                       0, STREAM, (const bf16*)x, (const bf16*)dy,
-                       (const bf16*)y, M, C, (const float*)mean,
-                       (const float*)invstd, (float*)sums2);
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(RED_BLOCKS),
+                       dim3(BLOCK), 0, STREAM, (const bf16*)x,
+                       (const bf16*)dy, (const bf16*)y, M, C,
+                       (const float*)mean, (const float*)invstd,
+                       (float*)partial);
+  return (int)hipGetLastError();
+}
+
+int bps_bn_fold(const void* partial, int C, void* sums2, void* stream) {
+  int blocks = (2 * C + BLOCK - 1) / BLOCK;
+  hipLaunchKernelGGL(bn_fold_kernel, dim3(blocks), dim3(BLOCK), 0, STREAM,
+                     (const float*)partial, C, (float*)sums2);
   return (int)hipGetLastError();
 }
 

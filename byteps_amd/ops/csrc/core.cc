@@ -57,8 +57,10 @@ int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
                      void* stream);
 int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
                       long long M, int C, const void* mean,
-                      const void* invstd, void* sums2, int relu,
+                      const void* invstd, void* partial, int relu,
                       void* stream);
+int bps_bn_fold(const void* partial, int C, void* sums2, void* stream);
+int bps_bn_red_blocks(void);
 int bps_bn_bwd_apply(const void* x, const void* dy, const void* y, void* dx,
                      void* dres, long long M, int C, const void* mean,
                      const void* invstd, const void* gamma, const void* sums2,
@@ -179,9 +181,14 @@ PYBIND11_MODULE(_core, m) {
         });
 
   // fused batchnorm
+  m.attr("BN_RED_BLOCKS") = bps_bn_red_blocks();
   m.def("bn_reduce", [](uintptr_t x, int64_t M, int C, uintptr_t sums,
                         uintptr_t s) {
     check(bps_bn_reduce(CP(x), M, C, P(sums), P(s)), "bps_bn_reduce");
+  });
+  m.def("bn_fold", [](uintptr_t partial, int C, uintptr_t sums2,
+                      uintptr_t s) {
+    check(bps_bn_fold(CP(partial), C, P(sums2), P(s)), "bps_bn_fold");
   });
   m.def("bn_finalize",
         [](uintptr_t sums, int64_t M, int C, float eps, float momentum,

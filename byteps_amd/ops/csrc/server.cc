@@ -117,7 +117,11 @@ struct Conn {
 struct InitPayload {
   uint64_t nelem;
   uint32_t expected;
-  uint32_t levels;  // dithering s
+  uint32_t levels;  // dithering s / sparse k
+  // optional extension (new workers send 24 bytes; 16-byte payloads get
+  // flags = 0)
+  uint32_t flags;   // bit0: server-side error feedback on the merged reply
+  uint32_t pad;
 };
 
 struct PendingPull {
@@ -127,7 +131,12 @@ struct PendingPull {
 
 struct KeyState {
   std::mutex mu;
-  std::vector<float> store;      // merged fp32 accumulator
+  std::vector<float> store;      // in-progress fp32 accumulator
+  std::vector<float> published;  // snapshot served to pulls (sync raw mode):
+                                 // without it a fast worker's next-round
+                                 // COPY overwrites the store while a slow
+                                 // worker's pull of the previous round is
+                                 // still outstanding
   std::vector<char> reply;       // compressed merged (codec mode)
   uint64_t nelem = 0;
   uint32_t expected = 1;
@@ -139,6 +148,9 @@ struct KeyState {
   std::vector<PendingPull> pending;
   uint64_t push_total = 0;       // scheduling signal
   std::vector<float> scratch;    // decompress workspace
+  bool server_ef = false;        // error-feedback on the merged reply
+  std::vector<float> ef_err;     // residual of the previous reply
+  std::vector<float> ef_comp;    // compensated merge workspace
 };
 
 struct Task {
@@ -330,9 +342,10 @@ class Server {
 
   void handle_init(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
                    const std::vector<char>& payload) {
-    if (payload.size() >= sizeof(InitPayload)) {
-      InitPayload ip;
-      std::memcpy(&ip, payload.data(), sizeof(ip));
+    if (payload.size() >= 16) {
+      InitPayload ip{};
+      std::memcpy(&ip, payload.data(),
+                  std::min(payload.size(), sizeof(ip)));
       std::lock_guard<std::mutex> lk(keys_mu_);
       auto& slot = keys_[h.key];
       if (!slot) {
@@ -342,6 +355,7 @@ class Server {
         slot->codec = cmd_codec(h.cmd);
         slot->levels = std::max(1u, ip.levels);
         slot->async_mode = cmd_async(h.cmd);
+        slot->server_ef = (ip.flags & 1u) != 0;
         // first-init page-aligned store (reference
         // server/server.cc:266-294); vector is 64 B aligned via resize
         slot->store.assign(ip.nelem, 0.0f);
@@ -384,8 +398,11 @@ class Server {
     r.op = kPullReply;
     r.aux = ks->version;
     if (ks->codec == kRaw) {
-      r.len = ks->store.size() * sizeof(float);
-      conn->send(r, ks->store.data());
+      const std::vector<float>& src =
+          (ks->async_mode || ks->published.empty()) ? ks->store
+                                                    : ks->published;
+      r.len = src.size() * sizeof(float);
+      conn->send(r, src.data());
     } else {
       r.len = ks->reply.size();
       conn->send(r, ks->reply.data());
@@ -470,7 +487,14 @@ class Server {
       // ALL_RECV: finalize merge (reference server/server.cc:348-370)
       ks->version++;
       ks->round_senders.clear();
-      if (ks->codec != kRaw) compress_reply(ks);
+      if (ks->codec != kRaw) {
+        compress_reply(ks);
+      } else {
+        // publish the merge; next round's COPY reuses the old buffer
+        ks->published.swap(ks->store);
+        if (ks->store.size() != ks->nelem)
+          ks->store.assign(ks->nelem, 0.0f);
+      }
       auto pending = std::move(ks->pending);
       ks->pending.clear();
       for (auto& p : pending) {
@@ -483,6 +507,19 @@ class Server {
   void compress_reply(KeyState* ks) {
     const int64_t n = (int64_t)ks->nelem;
     const float* acc = ks->store.data();
+    // server-side error feedback (reference server compressor mirror,
+    // server/server.cc:228-257 + vanilla EF): compensate the merge with
+    // the previous reply's residual before compressing, then store the
+    // new residual.
+    if (ks->server_ef) {
+      if (ks->ef_err.empty()) ks->ef_err.assign(n, 0.0f);
+      ks->ef_comp.resize(n);
+      const float* err = ks->ef_err.data();
+      float* comp = ks->ef_comp.data();
+#pragma omp parallel for
+      for (int64_t i = 0; i < n; ++i) comp[i] = acc[i] + err[i];
+      acc = comp;
+    }
     switch (ks->codec) {
       case kOnebit: {
         int64_t nwords = (n + 63) >> 6;
@@ -537,6 +574,44 @@ class Server {
       }
       default:
         break;
+    }
+
+    if (ks->server_ef) {
+      // new residual = compensated merge − decompress(reply)
+      ks->scratch.assign(n, 0.0f);
+      float* dec = ks->scratch.data();
+      switch (ks->codec) {
+        case kOnebit: {
+          int64_t nwords = (n + 63) >> 6;
+          float sc;
+          std::memcpy(&sc, ks->reply.data() + nwords * 8, 4);
+          bps_cpu_onebit_decompress((const uint64_t*)ks->reply.data(), sc, n,
+                                    dec);
+          break;
+        }
+        case kTopk:
+        case kRandomk: {
+          int64_t k = (int64_t)(ks->reply.size() / 8);
+          const int32_t* idx = (const int32_t*)ks->reply.data();
+          const float* val = (const float*)(ks->reply.data() + k * 4);
+          for (int64_t j = 0; j < k; ++j) dec[idx[j]] = val[j];
+          break;
+        }
+        case kDitherLinear:
+        case kDitherNatural: {
+          float norm;
+          std::memcpy(&norm, ks->reply.data(), 4);
+          bps_cpu_dithering_decompress(
+              (const int8_t*)(ks->reply.data() + 4), n, (int)ks->levels,
+              ks->codec == kDitherNatural, norm, dec);
+          break;
+        }
+        default:
+          break;
+      }
+      float* err = ks->ef_err.data();
+#pragma omp parallel for
+      for (int64_t i = 0; i < n; ++i) err[i] = acc[i] - dec[i];
     }
   }
 

@@ -44,11 +44,13 @@ class _FusedBNFunction(torch.autograd.Function):
         s = _stream(x)
         y = torch.empty_like(x)
         if training:
-            sums = torch.zeros(2 * C, dtype=torch.float32, device=dev)
-            core.bn_reduce(x.data_ptr(), M, C, sums.data_ptr(), s)
+            nb = core.BN_RED_BLOCKS
+            partial = torch.empty(nb * 2 * C, dtype=torch.float32,
+                                  device=dev)
+            core.bn_reduce(x.data_ptr(), M, C, partial.data_ptr(), s)
             mean = torch.empty(C, dtype=torch.float32, device=dev)
             invstd = torch.empty(C, dtype=torch.float32, device=dev)
-            core.bn_finalize(sums.data_ptr(), M, C, eps, momentum,
+            core.bn_finalize(partial.data_ptr(), M, C, eps, momentum,
                              mean.data_ptr(), invstd.data_ptr(),
                              running_mean.data_ptr(), running_var.data_ptr(),
                              1, s)
@@ -75,10 +77,14 @@ class _FusedBNFunction(torch.autograd.Function):
         core = K.core()
         dy = dy.contiguous(memory_format=torch.channels_last)
         s = _stream(x)
-        sums2 = torch.zeros(2 * C, dtype=torch.float32, device=x.device)
+        nb = core.BN_RED_BLOCKS
+        partial = torch.empty(nb * 2 * C, dtype=torch.float32,
+                              device=x.device)
         core.bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), y.data_ptr(), M, C,
                            mean.data_ptr(), invstd.data_ptr(),
-                           sums2.data_ptr(), int(ctx.relu), s)
+                           partial.data_ptr(), int(ctx.relu), s)
+        sums2 = torch.empty(2 * C, dtype=torch.float32, device=x.device)
+        core.bn_fold(partial.data_ptr(), C, sums2.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         core.bn_bwd_apply(x.data_ptr(), dy.data_ptr(), y.data_ptr(),

@@ -210,7 +210,12 @@ class PSPipeline:
         comp = ki.compressor
         codec = comp.codec if comp is not None else 0
         levels = comp.levels if comp is not None else 0
-        payload = struct.pack("<QII", ki.nelem, self.num_nodes, levels)
+        # bit0: ask the server to run error feedback on its merged reply
+        flags = 1 if (comp is not None and str(
+            self.compression_params.get("ef_type", "")).lower()
+            in ("vanilla", "1", "true")) else 0
+        payload = struct.pack("<QIIII", ki.nelem, self.num_nodes, levels,
+                              flags, 0)
         buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
         cmd = _make_cmd(codec, 0, self.cfg.enable_async)
         t = self.kv.submit(ki.server, _OP_INIT, ki.key, buf.data_ptr(),
