@@ -111,6 +111,21 @@ def main():
     if args.partition_mb:
         os.environ["BPS_PARTITION_BYTES"] = str(args.partition_mb * 2**20)
 
+    server = None
+    if args.compression != "none":
+        # BASELINE configs 4/5: compressed gradients through a colocated
+        # CPU PS server (port derived from MASTER_PORT so all ranks agree)
+        base_port = int(os.environ.get("MASTER_PORT", "29500"))
+        ps_port = base_port + 137
+        os.environ.setdefault("BPS_FORCE_DISTRIBUTED", "1")
+        os.environ.setdefault("BPS_NUM_SERVER", "1")
+        os.environ.setdefault("BPS_SERVER_URIS", "127.0.0.1:%d" % ps_port)
+        os.environ.setdefault("BPS_MIN_COMPRESS_BYTES", "65536")
+        if int(os.environ.get("LOCAL_RANK", "0")) == 0:
+            from byteps_amd.ops import core
+            server = core().Server(ps_port, 8, False)
+            server.start()
+
     import byteps_amd.torch as bps
     bps.init()
     world = bps.size()
@@ -123,8 +138,13 @@ def main():
 
     net, step_fn, per_step_items, unit, batch = build(args, device)
 
+    cparams = None
+    if args.compression != "none":
+        cparams = {"compressor_type": args.compression,
+                   "ef_type": "vanilla",
+                   "compressor_k": 4096}
     from byteps_amd.torch.parallel import DistributedDataParallel as DDP
-    model = DDP(net, broadcast_buffers=False)
+    model = DDP(net, broadcast_buffers=False, compression_params=cparams)
     opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
                           weight_decay=1e-4)
 
@@ -143,6 +163,8 @@ def main():
     # -- hipGraph capture ---------------------------------------------------
     graphed = None
     want_graph = args.graph == "on" or (args.graph == "auto" and on_gpu)
+    if args.compression != "none":
+        want_graph = False      # PS pipeline does host-side KV work per step
     if want_graph and on_gpu:
         try:
             for _ in range(3):      # warm up allocator + RCCL before capture
@@ -226,6 +248,8 @@ def main():
         }
         print(json.dumps(out))
     bps.shutdown()
+    if server is not None:
+        server.stop()
 
 
 if __name__ == "__main__":

@@ -87,8 +87,9 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
       }
     }
   }
-  // LDS reduce across groups for this c8 (sum pass, then sq pass)
-  float* out = partial + (long long)blockIdx.x * 2 * C;
+  // LDS reduce across groups for this c8 (sum pass, then sq pass).
+  // Partials are stored TRANSPOSED — partial[row][RED_BLOCKS] with
+  // row ∈ [0,2C) — so the folding kernels read coalesced along blocks.
 #pragma unroll
   for (int i = 0; i < 8; ++i) lds[t * 8 + i] = s.v[i];
   __syncthreads();
@@ -97,7 +98,8 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) out[(c8 << 3) + i] = s.v[i];
+    for (int i = 0; i < 8; ++i)
+      partial[(long long)((c8 << 3) + i) * RED_BLOCKS + blockIdx.x] = s.v[i];
   }
   __syncthreads();
 #pragma unroll
@@ -108,14 +110,28 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
 #pragma unroll
       for (int i = 0; i < 8; ++i) q.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) out[C + (c8 << 3) + i] = q.v[i];
+    for (int i = 0; i < 8; ++i)
+      partial[(long long)(C + (c8 << 3) + i) * RED_BLOCKS + blockIdx.x] =
+          q.v[i];
   }
+}
+
+// one-wave row reduction of a [RED_BLOCKS]-wide partial row
+__device__ inline float wave_row_sum(const float* __restrict__ row) {
+  const int lane = threadIdx.x & 63;
+  float acc = 0.0f;
+#pragma unroll
+  for (int b = lane; b < RED_BLOCKS; b += 64) acc += row[b];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  return acc;  // valid in lane 0
 }
 
 // ---------------------------------------------------------------------------
 // finalize: mean/invstd from sums; running-stat update (1 thread/channel)
 // ---------------------------------------------------------------------------
 
+// one wave per channel: lanes stride the transposed partial rows
 __global__ void bn_finalize_kernel(const float* __restrict__ partial,
                                    long long M, int C, float eps,
                                    float momentum,
@@ -124,13 +140,11 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partial,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
                                    int update_running) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int c = blockIdx.x;
   if (c >= C) return;
-  float sum = 0.0f, sumsq = 0.0f;
-  for (int b = 0; b < RED_BLOCKS; ++b) {
-    sum += partial[(long long)b * 2 * C + c];
-    sumsq += partial[(long long)b * 2 * C + C + c];
-  }
+  float sum = wave_row_sum(partial + (long long)c * RED_BLOCKS);
+  float sumsq = wave_row_sum(partial + (long long)(C + c) * RED_BLOCKS);
+  if (threadIdx.x != 0) return;
   float n = (float)M;
   float mean = sum / n;
   float var = sumsq / n - mean * mean;
@@ -244,15 +258,13 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
   }
 }
 
-// fold the RED_BLOCKS partials into sums2[2C]
+// fold the transposed partials into sums2[2C] — one wave per row
 __global__ void bn_fold_kernel(const float* __restrict__ partial, int C,
                                float* __restrict__ sums2) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= 2 * C) return;
-  float acc = 0.0f;
-  for (int b = 0; b < RED_BLOCKS; ++b)
-    acc += partial[(long long)b * 2 * C + c];
-  sums2[c] = acc;
+  int row = blockIdx.x;
+  if (row >= 2 * C) return;
+  float acc = wave_row_sum(partial + (long long)row * RED_BLOCKS);
+  if (threadIdx.x == 0) sums2[row] = acc;
 }
 
 // ---------------------------------------------------------------------------
@@ -327,8 +339,7 @@ int bps_bn_finalize(const void* sums, long long M, int C, float eps,
                     float momentum, void* mean_out, void* invstd_out,
                     void* running_mean, void* running_var, int update_running,
                     void* stream) {
-  int blocks = (C + BLOCK - 1) / BLOCK;
-  hipLaunchKernelGGL(bn_finalize_kernel, dim3(blocks), dim3(BLOCK), 0,
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(C), dim3(64), 0,
                      STREAM, (const float*)sums, M, C, eps, momentum,
                      (float*)mean_out, (float*)invstd_out,
                      (float*)running_mean, (float*)running_var,
@@ -376,8 +387,7 @@ int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
 }
 
 int bps_bn_fold(const void* partial, int C, void* sums2, void* stream) {
-  int blocks = (2 * C + BLOCK - 1) / BLOCK;
-  hipLaunchKernelGGL(bn_fold_kernel, dim3(blocks), dim3(BLOCK), 0, STREAM,
+  hipLaunchKernelGGL(bn_fold_kernel, dim3(2 * C), dim3(64), 0, STREAM,
                      (const float*)partial, C, (float*)sums2);
   return (int)hipGetLastError();
 }
