@@ -16,6 +16,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..torch.fused_ln import FusedLayerNorm
+
 
 class BertConfig:
     def __init__(self, vocab_size=30522, hidden=1024, layers=24, heads=16,
@@ -65,10 +67,10 @@ class EncoderLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.attn = SelfAttention(cfg)
-        self.ln1 = nn.LayerNorm(cfg.hidden)
+        self.ln1 = FusedLayerNorm(cfg.hidden)
         self.fc1 = nn.Linear(cfg.hidden, cfg.intermediate)
         self.fc2 = nn.Linear(cfg.intermediate, cfg.hidden)
-        self.ln2 = nn.LayerNorm(cfg.hidden)
+        self.ln2 = FusedLayerNorm(cfg.hidden)
 
     def forward(self, x):
         x = self.ln1(x + self.attn(x))
@@ -83,11 +85,11 @@ class BertForPreTraining(nn.Module):
         self.tok_emb = nn.Embedding(cfg.vocab_size, cfg.hidden)
         self.pos_emb = nn.Embedding(cfg.max_pos, cfg.hidden)
         self.type_emb = nn.Embedding(2, cfg.hidden)
-        self.emb_ln = nn.LayerNorm(cfg.hidden)
+        self.emb_ln = FusedLayerNorm(cfg.hidden)
         self.layers = nn.ModuleList(
             EncoderLayer(cfg) for _ in range(cfg.layers))
         self.mlm_dense = nn.Linear(cfg.hidden, cfg.hidden)
-        self.mlm_ln = nn.LayerNorm(cfg.hidden)
+        self.mlm_ln = FusedLayerNorm(cfg.hidden)
         # decoder tied to token embedding
         self.mlm_bias = nn.Parameter(torch.zeros(cfg.vocab_size))
         self.apply(self._init)
@@ -105,6 +107,11 @@ class BertForPreTraining(nn.Module):
         if token_type_ids is not None:
             x = x + self.type_emb(token_type_ids)
         x = self.emb_ln(x)
+        # keep the residual stream in the autocast compute dtype so the
+        # fused bf16 LayerNorm kernels engage and per-layer re-casts
+        # disappear (profiles/bert_large_steady_state.md)
+        if x.is_cuda and torch.is_autocast_enabled():
+            x = x.to(torch.get_autocast_gpu_dtype())
         for layer in self.layers:
             x = layer(x)
         h = self.mlm_ln(F.gelu(self.mlm_dense(x), approximate="tanh"))

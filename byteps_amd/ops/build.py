@@ -14,7 +14,7 @@ CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "_core.so")
 
 SOURCES = ["core.cc", "kv.cc", "server.cc", "cpu_reducer.cc",
-           "kernels.hip", "compress.hip", "bn.hip"]
+           "kernels.hip", "compress.hip", "bn.hip", "ln.hip"]
 
 
 def _newer_than_out(paths) -> bool:

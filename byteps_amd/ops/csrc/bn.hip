@@ -234,7 +234,7 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
       }
     }
   }
-  float* out = partial + (long long)blockIdx.x * 2 * C;
+  // transposed partial layout [row][RED_BLOCKS], same as the fwd reduce
 #pragma unroll
   for (int i = 0; i < 8; ++i) lds[t * 8 + i] = s1.v[i];
   __syncthreads();
@@ -243,7 +243,9 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s1.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) out[(c8 << 3) + i] = s1.v[i];
+    for (int i = 0; i < 8; ++i)
+      partial[(long long)((c8 << 3) + i) * RED_BLOCKS + blockIdx.x] =
+          s1.v[i];
   }
   __syncthreads();
 #pragma unroll
@@ -254,7 +256,9 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
 #pragma unroll
       for (int i = 0; i < 8; ++i) s2.v[i] += lds[(gg * cpt + c8) * 8 + i];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) out[C + (c8 << 3) + i] = s2.v[i];
+    for (int i = 0; i < 8; ++i)
+      partial[(long long)(C + (c8 << 3) + i) * RED_BLOCKS + blockIdx.x] =
+          s2.v[i];
   }
 }
 

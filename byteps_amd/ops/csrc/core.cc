@@ -61,6 +61,17 @@ int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
                       void* stream);
 int bps_bn_fold(const void* partial, int C, void* sums2, void* stream);
 int bps_bn_red_blocks(void);
+
+// -- fused layernorm (ln.hip) ----------------------------------------------
+int bps_ln_supported(int C);
+int bps_ln_red_blocks(void);
+int bps_ln_fwd(const void* x, const void* gamma, const void* beta, void* y,
+               long long M, int C, float eps, void* mean, void* invstd,
+               void* stream);
+int bps_ln_bwd(const void* x, const void* dy, const void* gamma,
+               const void* mean, const void* invstd, void* dx, long long M,
+               int C, void* partial, void* stream);
+int bps_ln_fold(const void* partial, int C, void* sums2, void* stream);
 int bps_bn_bwd_apply(const void* x, const void* dy, const void* y, void* dx,
                      void* dres, long long M, int C, const void* mean,
                      const void* invstd, const void* gamma, const void* sums2,
@@ -223,6 +234,30 @@ PYBIND11_MODULE(_core, m) {
                                  relu, P(s)),
                 "bps_bn_bwd_apply");
         });
+
+  // fused layernorm
+  m.attr("LN_RED_BLOCKS") = bps_ln_red_blocks();
+  m.def("ln_supported", [](int C) { return bps_ln_supported(C) != 0; });
+  m.def("ln_fwd",
+        [](uintptr_t x, uintptr_t gamma, uintptr_t beta, uintptr_t y,
+           int64_t M, int C, float eps, uintptr_t mean, uintptr_t invstd,
+           uintptr_t s) {
+          check(bps_ln_fwd(CP(x), CP(gamma), CP(beta), P(y), M, C, eps,
+                           P(mean), P(invstd), P(s)),
+                "bps_ln_fwd");
+        });
+  m.def("ln_bwd",
+        [](uintptr_t x, uintptr_t dy, uintptr_t gamma, uintptr_t mean,
+           uintptr_t invstd, uintptr_t dx, int64_t M, int C,
+           uintptr_t partial, uintptr_t s) {
+          check(bps_ln_bwd(CP(x), CP(dy), CP(gamma), CP(mean), CP(invstd),
+                           P(dx), M, C, P(partial), P(s)),
+                "bps_ln_bwd");
+        });
+  m.def("ln_fold", [](uintptr_t partial, int C, uintptr_t sums2,
+                      uintptr_t s) {
+    check(bps_ln_fold(CP(partial), C, P(sums2), P(s)), "bps_ln_fold");
+  });
 
   // CPU reducer / codecs
   m.def("cpu_sum", [](uintptr_t d, uintptr_t s, int64_t n, int dt) {
