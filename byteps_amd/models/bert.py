@@ -44,6 +44,21 @@ class BertConfig:
                           intermediate=128, max_pos=128)
 
 
+def _sdpa_ctx():
+    """Optional backend pin via BPS_SDPA_BACKEND=flash|efficient|math —
+    lets short-sequence configs A/B the attention backward (flash bwd is
+    overhead-bound at s=128 on ROCm)."""
+    import os
+    name = os.environ.get("BPS_SDPA_BACKEND", "").lower()
+    if not name:
+        return None
+    from torch.nn.attention import SDPBackend, sdpa_kernel
+    table = {"flash": SDPBackend.FLASH_ATTENTION,
+             "efficient": SDPBackend.EFFICIENT_ATTENTION,
+             "math": SDPBackend.MATH}
+    return sdpa_kernel(table[name]) if name in table else None
+
+
 class SelfAttention(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
@@ -57,8 +72,15 @@ class SelfAttention(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
         q, k, v = qkv.permute(2, 0, 3, 1, 4)          # 3 × (B, h, S, d)
-        o = F.scaled_dot_product_attention(
-            q, k, v, dropout_p=self.dropout if self.training else 0.0)
+        ctx = _sdpa_ctx()
+        if ctx is not None:
+            with ctx:
+                o = F.scaled_dot_product_attention(
+                    q, k, v,
+                    dropout_p=self.dropout if self.training else 0.0)
+        else:
+            o = F.scaled_dot_product_attention(
+                q, k, v, dropout_p=self.dropout if self.training else 0.0)
         o = o.transpose(1, 2).reshape(B, S, H)
         return self.out(o)
 
