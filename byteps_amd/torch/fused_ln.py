@@ -74,5 +74,8 @@ class FusedLayerNorm(nn.Module):
                 and K.have_core() and K.core().ln_supported(self.hidden)):
             return _FusedLNFunction.apply(x, self.weight, self.bias,
                                           self.eps)
+        if x.dtype != self.weight.dtype:
+            return F.layer_norm(x.float(), (self.hidden,), self.weight,
+                                self.bias, self.eps).to(x.dtype)
         return F.layer_norm(x, (self.hidden,), self.weight, self.bias,
                             self.eps)
