@@ -44,14 +44,16 @@ class BertConfig:
                           intermediate=128, max_pos=128)
 
 
-def _sdpa_ctx():
-    """Optional backend pin via BPS_SDPA_BACKEND=flash|efficient|math —
-    lets short-sequence configs A/B the attention backward (flash bwd is
-    overhead-bound at s=128 on ROCm)."""
+def _sdpa_ctx(seq_len: int):
+    """SDPA backend policy: flash backward is overhead-bound at short
+    sequences on ROCm (measured 140 µs vs 37 µs fwd at s=128 —
+    profiles/bert_large_steady_state.md), so default to the CK
+    efficient-attention kernels for S ≤ 256 and flash beyond.
+    ``BPS_SDPA_BACKEND=flash|efficient|math`` overrides."""
     import os
     name = os.environ.get("BPS_SDPA_BACKEND", "").lower()
     if not name:
-        return None
+        name = "efficient" if seq_len <= 256 else "flash"
     from torch.nn.attention import SDPBackend, sdpa_kernel
     table = {"flash": SDPBackend.FLASH_ATTENTION,
              "efficient": SDPBackend.EFFICIENT_ATTENTION,
@@ -72,7 +74,7 @@ class SelfAttention(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
         q, k, v = qkv.permute(2, 0, 3, 1, 4)          # 3 × (B, h, S, d)
-        ctx = _sdpa_ctx()
+        ctx = _sdpa_ctx(S) if x.is_cuda else None
         if ctx is not None:
             with ctx:
                 o = F.scaled_dot_product_attention(

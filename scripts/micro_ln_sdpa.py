@@ -1,7 +1,11 @@
 """Microbench on MI355X: fused LN vs torch LN (autocast), and SDPA
 backend comparison at BERT-large shapes (b64 s128 h16 d64)."""
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 import torch.nn.functional as F
