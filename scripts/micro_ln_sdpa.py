@@ -47,6 +47,35 @@ def main():
     print("LN fused fwd+bwd: %.1f us" % timeit(fused_fb))
     print("LN torch fwd+bwd: %.1f us" % timeit(torch_fb))
 
+    # kernel-level (no autograd engine): forward only under no_grad
+    from byteps_amd import ops as K
+    core = K.core()
+    xk = x16.detach()
+    yk = torch.empty_like(xk)
+    mean = torch.empty(M, device="cuda")
+    invstd = torch.empty(M, device="cuda")
+    wf = fused.weight.detach().float().contiguous()
+    bf = fused.bias.detach().float().contiguous()
+    s = torch.cuda.current_stream().cuda_stream
+
+    def fused_fwd_kernel():
+        core.ln_fwd(xk.data_ptr(), wf.data_ptr(), bf.data_ptr(),
+                    yk.data_ptr(), M, C, 1e-12, mean.data_ptr(),
+                    invstd.data_ptr(), s)
+
+    def torch_fwd_kernel():
+        with torch.no_grad():
+            F.layer_norm(xk.float(), (C,), wf, bf, 1e-12)
+
+    def torch_fwd_bf16():
+        with torch.no_grad():
+            F.layer_norm(xk, (C,), wf.to(torch.bfloat16),
+                         bf.to(torch.bfloat16), 1e-12)
+
+    print("LN fused fwd kernel:      %.1f us" % timeit(fused_fwd_kernel, 200))
+    print("LN torch fwd fp32(+cast): %.1f us" % timeit(torch_fwd_kernel, 200))
+    print("LN torch fwd bf16:        %.1f us" % timeit(torch_fwd_bf16, 200))
+
     # SDPA backends
     B, h, S, d = 64, 16, 128, 64
     from torch.nn.attention import SDPBackend, sdpa_kernel
