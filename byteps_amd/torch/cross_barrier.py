@@ -119,6 +119,8 @@ class CrossBarrier:
             eng._ps.wait(bucket.ps_ticket)
         elif bucket.work is not None:
             bucket.work.wait()
+            if eng.comm_dtype is not None:
+                bucket.buffer.copy_(eng._wire_scratch[bucket.plan.index])
         if eng.average and eng.world > 1 and not eng.prescale:
             bucket.buffer.div_(eng.world)
         # refresh live hyperparams (lr schedules) from the user optimizer

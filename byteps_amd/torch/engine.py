@@ -203,6 +203,19 @@ class GradEngine:
                 p.grad = torch.zeros_like(p)
                 self._split_params[pidx] = bks
 
+        # optional reduced-precision wire for the RCCL collectives
+        # (BPS_COMM_DTYPE=bf16): cast bucket → persistent bf16 scratch,
+        # all-reduce the scratch, cast back at synchronize.  Halves xGMI
+        # bytes with bf16's fp32 exponent range — the role the
+        # reference's fp16 Compression played (torch/compression.py).
+        from ..common.config import env_str
+        wire = env_str("BPS_COMM_DTYPE", default="").lower()
+        self.comm_dtype = torch.bfloat16 if wire in ("bf16", "bfloat16") \
+            else torch.float16 if wire in ("fp16", "half") else None
+        if self.comm_dtype is not None and dt == self.comm_dtype:
+            self.comm_dtype = None
+        self._wire_scratch: Dict[int, torch.Tensor] = {}
+
         self._ps = None
         if C._state.ps_enabled:
             from . import ps_pipeline
@@ -284,6 +297,16 @@ class GradEngine:
                 b.buffer.div_(self.world)
             if self._ps is not None:
                 b.ps_ticket = self._ps.submit(b)
+            elif self.comm_dtype is not None:
+                scratch = self._wire_scratch.get(b.plan.index)
+                if scratch is None:
+                    scratch = torch.empty_like(b.buffer,
+                                               dtype=self.comm_dtype)
+                    self._wire_scratch[b.plan.index] = scratch
+                scratch.copy_(b.buffer)
+                b.work = dist.all_reduce(
+                    scratch, op=dist.ReduceOp.SUM, group=self.group,
+                    async_op=True)
             else:
                 b.work = dist.all_reduce(
                     b.buffer, op=dist.ReduceOp.SUM, group=self.group,
@@ -340,6 +363,8 @@ class GradEngine:
                 self._ps.wait(b.ps_ticket)
             elif b.work is not None:
                 b.work.wait()
+                if self.comm_dtype is not None:
+                    b.buffer.copy_(self._wire_scratch[b.plan.index])
         if self.average and self.world > 1 and not self.prescale:
             torch._foreach_div_([b.buffer for b in self.buckets],
                                 float(self.world))

@@ -136,3 +136,34 @@ def test_param_split_across_partitions():
         _split_param, 2, extra_env={"BPS_PARTITION_BYTES": "32768"})
     for p0, p1 in zip(*results):
         assert torch.allclose(p0, p1, rtol=1e-5, atol=1e-6)
+
+
+def _comm_bf16(rank, world):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    opt = bps.DistributedOptimizer(opt, named_parameters=m.named_parameters())
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    for _ in range(3):
+        opt.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_comm_dtype_bf16_wire():
+    """BPS_COMM_DTYPE=bf16: collectives run on a bf16 scratch; results
+    must match fp32 within bf16 tolerance and agree across ranks."""
+    world = 2
+    results = run_in_processes(_comm_bf16, world,
+                               extra_env={"BPS_COMM_DTYPE": "bf16"})
+    ref = run_in_processes(_comm_bf16, world)
+    for p_a, p_b in zip(*results):
+        assert torch.allclose(p_a, p_b)     # ranks agree exactly
+    for p_got, p_ref in zip(results[0], ref[0]):
+        assert torch.allclose(p_got, p_ref, rtol=3e-2, atol=3e-2)
