@@ -148,6 +148,21 @@ def main():
     opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
                           weight_decay=1e-4)
 
+    # pre-tuned hipBLASLt algorithm selections for gfx950 (TunableOp,
+    # generated offline by scripts/gen_tunableop.py — tuning is too slow
+    # to run inside a timed benchmark)
+    tuning_file = os.path.join(
+        os.path.dirname(os.path.abspath(__file__)),
+        "byteps_amd", "tuning", "tunableop_gfx950.csv")
+    if on_gpu and os.path.exists(tuning_file):
+        try:
+            torch.cuda.tunable.enable(True)
+            torch.cuda.tunable.tuning_enable(False)
+            torch.cuda.tunable.set_filename(tuning_file)
+            torch.cuda.tunable.read_file()
+        except Exception as e:
+            print("tunableop load failed: %s" % e, file=sys.stderr)
+
     use_bf16 = args.dtype == "bf16" and on_gpu
     autocast = torch.autocast("cuda", dtype=torch.bfloat16) if use_bf16 \
         else torch.autocast("cpu", enabled=False)
