@@ -349,9 +349,12 @@ class GradEngine:
             for b in self.buckets:
                 if not b.issued:
                     heapq.heappush(self._pending, (-b.priority, b.plan.index))
-            self._credit_save, self._credit = getattr(self, "_credit", 0), 0
-            self._drain_locked()
-            self._credit = self._credit_save
+            saved_credit = self._credit
+            self._credit = 0          # credits never hold back a flush
+            try:
+                self._drain_locked()
+            finally:
+                self._credit = saved_credit
 
     def synchronize(self) -> None:
         """Wait for all issued buckets; apply deferred averaging; reset."""

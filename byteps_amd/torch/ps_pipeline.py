@@ -241,9 +241,7 @@ class PSPipeline:
                 # non-root: nothing to push; wait() broadcasts the result
                 fut = Future()
                 fut.set_result((None, None))
-                t = Ticket(fut, bucket, buf)
-                t.reply_view = ("bcast", root_global)
-                return t
+                return Ticket(fut, bucket, buf)
             shard = buf
         elif self.node_world > 1:
             per = buf.numel() // self.node_world
@@ -288,7 +286,6 @@ class PSPipeline:
                 st.stream.wait_event(rs_event)
             shard_f = shard if shard.dtype == torch.float32 \
                 else shard.float()
-            aux = ki.round
             if comp is not None:
                 cp = comp.compress(shard_f)
                 payload = torch.cat(
