@@ -157,8 +157,13 @@ class GradEngine:
         # reference (mxnet/__init__.py:58-60, scheduled_queue.cc:86-95).
         rev = list(range(len(self.params)))[::-1]
         sizes = [self.params[i].numel() for i in rev]
-        # align so reduce-scatter shards divide evenly for any world ≤ 64
-        plans = plan_partitions(sizes, part_elems, align=64)
+        # align so reduce-scatter shards divide evenly for THIS world:
+        # lcm(64, world) keeps 16-byte vector alignment for the HIP
+        # kernels and exact shard division for any rank count (a plain 64
+        # silently truncates the bucket tail at e.g. world=3 or 6)
+        import math
+        align = 64 * self.world // math.gcd(64, self.world)
+        plans = plan_partitions(sizes, part_elems, align=align)
 
         self.buckets: List[Bucket] = []
         self.param_bucket: Dict[int, List[Bucket]] = {}   # param idx → buckets

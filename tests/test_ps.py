@@ -237,3 +237,14 @@ def test_ps_reduce_roots_parity(server):
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def test_ps_three_ranks_one_node_parity(server):
+    """Non-power-of-two world: bucket alignment must make shards divide
+    exactly (a fixed align=64 silently dropped the tail at world=3)."""
+    expected = _baseline(3, 3)
+    results = run_in_processes(_ps_worker, 3, 3, None,
+                               extra_env=_ps_env(server))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
