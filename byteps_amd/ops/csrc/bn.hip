@@ -162,6 +162,9 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partial,
 // fwd apply: y = [relu]( (x-mean)*invstd*gamma + beta [+ res] )
 // ---------------------------------------------------------------------------
 
+// With RELU the kernel also emits a 1-bit activation mask (one byte per
+// 8-channel slot) so the backward never re-reads y — 16 B of y becomes
+// 1 B of mask on the backward's critical path.
 template <bool RELU, bool RES>
 __global__ void bn_fwd_apply_kernel(const bf16* __restrict__ x,
                                     const bf16* __restrict__ res,
@@ -169,7 +172,8 @@ __global__ void bn_fwd_apply_kernel(const bf16* __restrict__ x,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ invstd,
                                     const float* __restrict__ gamma,
-                                    const float* __restrict__ beta) {
+                                    const float* __restrict__ beta,
+                                    unsigned char* __restrict__ mask) {
   const int cpt = C >> 3;
   const long long total = M * cpt;
   for (long long idx = (long long)blockIdx.x * BLOCK + threadIdx.x;
@@ -181,15 +185,20 @@ __global__ void bn_fwd_apply_kernel(const bf16* __restrict__ x,
     float vals[8], rv[8];
     load8(x + off, vals);
     if (RES) load8(res + off, rv);
+    unsigned char mbits = 0;
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       float scale = gamma[c0 + i] * invstd[c0 + i];
       float v = (vals[i] - mean[c0 + i]) * scale + beta[c0 + i];
       if (RES) v += rv[i];
-      if (RELU) v = v > 0.0f ? v : 0.0f;
+      if (RELU) {
+        if (v > 0.0f) mbits |= (unsigned char)(1u << i);
+        else v = 0.0f;
+      }
       vals[i] = v;
     }
     store8(y + off, vals);
+    if (RELU) mask[idx] = mbits;
   }
 }
 
@@ -202,7 +211,8 @@ __global__ void bn_fwd_apply_kernel(const bf16* __restrict__ x,
 template <bool RELU>
 __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
                                      const bf16* __restrict__ dy,
-                                     const bf16* __restrict__ y, long long M,
+                                     const unsigned char* __restrict__ mask,
+                                     long long M,
                                      int C, const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ partial) {
@@ -217,17 +227,17 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
   s1.zero();
   s2.zero();
   if (g < groups) {
-    float xv[8], dv[8], yv[8];
+    float xv[8], dv[8];
     for (long long row = (long long)blockIdx.x * groups + g; row < M;
          row += (long long)gridDim.x * groups) {
       const long long off = row * C + (c8 << 3);
       load8(x + off, xv);
       load8(dy + off, dv);
-      if (RELU) load8(y + off, yv);
+      const unsigned char mbits = RELU ? mask[row * cpt + c8] : 0;
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
         const int c = (c8 << 3) + i;
-        float dz = RELU ? (yv[i] > 0.0f ? dv[i] : 0.0f) : dv[i];
+        float dz = RELU ? ((mbits >> i) & 1 ? dv[i] : 0.0f) : dv[i];
         float xhat = (xv[i] - mean[c]) * invstd[c];
         s1.v[i] += dz;
         s2.v[i] += dz * xhat;
@@ -278,7 +288,7 @@ __global__ void bn_fold_kernel(const float* __restrict__ partial, int C,
 template <bool RELU, bool RES>
 __global__ void bn_bwd_apply_kernel(const bf16* __restrict__ x,
                                     const bf16* __restrict__ dy,
-                                    const bf16* __restrict__ y,
+                                    const unsigned char* __restrict__ mask,
                                     bf16* __restrict__ dx,
                                     bf16* __restrict__ dres, long long M,
                                     int C, const float* __restrict__ mean,
@@ -294,14 +304,14 @@ __global__ void bn_bwd_apply_kernel(const bf16* __restrict__ x,
     const int c8 = (int)(idx - row * cpt);
     const int c0 = c8 << 3;
     const long long off = row * C + c0;
-    float xv[8], dv[8], yv[8], dzv[8];
+    float xv[8], dv[8], dzv[8];
     load8(x + off, xv);
     load8(dy + off, dv);
-    if (RELU) load8(y + off, yv);
+    const unsigned char mbits = RELU ? mask[idx] : 0;
 #pragma unroll
     for (int i = 0; i < 8; ++i) {
       const int c = c0 + i;
-      float dz = RELU ? (yv[i] > 0.0f ? dv[i] : 0.0f) : dv[i];
+      float dz = RELU ? ((mbits >> i) & 1 ? dv[i] : 0.0f) : dv[i];
       dzv[i] = dz;
       float xhat = (xv[i] - mean[c]) * invstd[c];
       dv[i] = gamma[c] * invstd[c] *
@@ -354,14 +364,16 @@ int bps_bn_finalize(const void* sums, long long M, int C, float eps,
 int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
                      int C, const void* mean, const void* invstd,
                      const void* gamma, const void* beta, int relu,
-                     void* stream) {
+                     void* mask, void* stream) {
   if ((C & 7) || C > 2048) return -1;
+  if (relu && !mask) return -2;
   int g = grid_for_elems(M * (C >> 3));
 #define LAUNCH_FWD(R, S)                                                    \
   hipLaunchKernelGGL((bn_fwd_apply_kernel<R, S>), dim3(g), dim3(BLOCK), 0,  \
                      STREAM, (const bf16*)x, (const bf16*)res, (bf16*)y, M, \
                      C, (const float*)mean, (const float*)invstd,           \
-                     (const float*)gamma, (const float*)beta)
+                     (const float*)gamma, (const float*)beta,               \
+                     (unsigned char*)mask)
   if (relu && res) LAUNCH_FWD(true, true);
   else if (relu) LAUNCH_FWD(true, false);
   else if (res) LAUNCH_FWD(false, true);
@@ -370,21 +382,22 @@ int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
   return (int)hipGetLastError();
 }
 
-int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
+int bps_bn_bwd_reduce(const void* x, const void* dy, const void* mask,
                       long long M, int C, const void* mean,
                       const void* invstd, void* partial, int relu,
                       void* stream) {
   if ((C & 7) || C > 2048) return -1;
+  if (relu && !mask) return -2;
   if (relu)
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(RED_BLOCKS),
                        dim3(BLOCK), 0, STREAM, (const bf16*)x,
-                       (const bf16*)dy, (const bf16*)y, M, C,
+                       (const bf16*)dy, (const unsigned char*)mask, M, C,
                        (const float*)mean, (const float*)invstd,
                        (float*)partial);
   else
     hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(RED_BLOCKS),
                        dim3(BLOCK), 0, STREAM, (const bf16*)x,
-                       (const bf16*)dy, (const bf16*)y, M, C,
+                       (const bf16*)dy, (const unsigned char*)mask, M, C,
                        (const float*)mean, (const float*)invstd,
                        (float*)partial);
   return (int)hipGetLastError();
@@ -396,16 +409,19 @@ int bps_bn_fold(const void* partial, int C, void* sums2, void* stream) {
   return (int)hipGetLastError();
 }
 
-int bps_bn_bwd_apply(const void* x, const void* dy, const void* y, void* dx,
+int bps_bn_bwd_apply(const void* x, const void* dy, const void* mask,
+                     void* dx,
                      void* dres, long long M, int C, const void* mean,
                      const void* invstd, const void* gamma, const void* sums2,
                      int relu, void* stream) {
   if ((C & 7) || C > 2048) return -1;
+  if (relu && !mask) return -2;
   int g = grid_for_elems(M * (C >> 3));
 #define LAUNCH_BWD(R, S)                                                     \
   hipLaunchKernelGGL((bn_bwd_apply_kernel<R, S>), dim3(g), dim3(BLOCK), 0,   \
                      STREAM, (const bf16*)x, (const bf16*)dy,                \
-                     (const bf16*)y, (bf16*)dx, (bf16*)dres, M, C,           \
+                     (const unsigned char*)mask, (bf16*)dx, (bf16*)dres,     \
+                     M, C,                                                   \
                      (const float*)mean, (const float*)invstd,               \
                      (const float*)gamma, (const float*)sums2)
   if (relu && dres) LAUNCH_BWD(true, true);

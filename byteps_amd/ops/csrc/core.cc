@@ -54,8 +54,8 @@ int bps_bn_finalize(const void* sums, long long M, int C, float eps,
 int bps_bn_fwd_apply(const void* x, const void* res, void* y, long long M,
                      int C, const void* mean, const void* invstd,
                      const void* gamma, const void* beta, int relu,
-                     void* stream);
-int bps_bn_bwd_reduce(const void* x, const void* dy, const void* y,
+                     void* mask, void* stream);
+int bps_bn_bwd_reduce(const void* x, const void* dy, const void* mask,
                       long long M, int C, const void* mean,
                       const void* invstd, void* partial, int relu,
                       void* stream);
@@ -72,7 +72,8 @@ int bps_ln_bwd(const void* x, const void* dy, const void* gamma,
                const void* mean, const void* invstd, void* dx, long long M,
                int C, void* partial, void* stream);
 int bps_ln_fold(const void* partial, int C, void* sums2, void* stream);
-int bps_bn_bwd_apply(const void* x, const void* dy, const void* y, void* dx,
+int bps_bn_bwd_apply(const void* x, const void* dy, const void* mask,
+                     void* dx,
                      void* dres, long long M, int C, const void* mean,
                      const void* invstd, const void* gamma, const void* sums2,
                      int relu, void* stream);
@@ -212,9 +213,10 @@ PYBIND11_MODULE(_core, m) {
   m.def("bn_fwd_apply",
         [](uintptr_t x, uintptr_t res, uintptr_t y, int64_t M, int C,
            uintptr_t mean, uintptr_t invstd, uintptr_t gamma, uintptr_t beta,
-           int relu, uintptr_t s) {
+           int relu, uintptr_t mask, uintptr_t s) {
           check(bps_bn_fwd_apply(CP(x), CP(res), P(y), M, C, CP(mean),
-                                 CP(invstd), CP(gamma), CP(beta), relu, P(s)),
+                                 CP(invstd), CP(gamma), CP(beta), relu,
+                                 P(mask), P(s)),
                 "bps_bn_fwd_apply");
         });
   m.def("bn_bwd_reduce",

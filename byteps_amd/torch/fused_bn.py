@@ -59,12 +59,17 @@ class _FusedBNFunction(torch.autograd.Function):
             invstd = torch.rsqrt(running_var.float() + eps)
         wf = weight.float().contiguous()
         bf = bias.float().contiguous()
+        # relu path emits a 1-bit activation mask (1 byte / 8 channels) so
+        # backward never re-reads y
+        mask = torch.empty(M * (C // 8), dtype=torch.uint8, device=dev) \
+            if relu else None
         core.bn_fwd_apply(x.data_ptr(),
                           residual.data_ptr() if residual is not None else 0,
                           y.data_ptr(), M, C, mean.data_ptr(),
                           invstd.data_ptr(), wf.data_ptr(), bf.data_ptr(),
-                          int(relu), s)
-        ctx.save_for_backward(x, y, mean, invstd, wf)
+                          int(relu), mask.data_ptr() if relu else 0, s)
+        ctx.save_for_backward(x, mean, invstd, wf,
+                              mask if mask is not None else x.new_empty(0))
         ctx.relu = relu
         ctx.has_res = residual is not None
         ctx.dims = (M, C)
@@ -72,22 +77,23 @@ class _FusedBNFunction(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, invstd, wf = ctx.saved_tensors
+        x, mean, invstd, wf, mask = ctx.saved_tensors
         M, C = ctx.dims
         core = K.core()
         dy = dy.contiguous(memory_format=torch.channels_last)
         s = _stream(x)
         nb = core.BN_RED_BLOCKS
+        mask_ptr = mask.data_ptr() if ctx.relu else 0
         partial = torch.empty(nb * 2 * C, dtype=torch.float32,
                               device=x.device)
-        core.bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), y.data_ptr(), M, C,
+        core.bn_bwd_reduce(x.data_ptr(), dy.data_ptr(), mask_ptr, M, C,
                            mean.data_ptr(), invstd.data_ptr(),
                            partial.data_ptr(), int(ctx.relu), s)
         sums2 = torch.empty(2 * C, dtype=torch.float32, device=x.device)
         core.bn_fold(partial.data_ptr(), C, sums2.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
-        core.bn_bwd_apply(x.data_ptr(), dy.data_ptr(), y.data_ptr(),
+        core.bn_bwd_apply(x.data_ptr(), dy.data_ptr(), mask_ptr,
                           dx.data_ptr(),
                           dres.data_ptr() if dres is not None else 0, M, C,
                           mean.data_ptr(), invstd.data_ptr(), wf.data_ptr(),
@@ -116,6 +122,9 @@ class FusedBNReLU(nn.Module):
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
+        # kept for state_dict parity with nn.BatchNorm2d but NOT bumped
+        # per step on the GPU path — with fixed momentum it is unused,
+        # and the per-call long-add kernel showed up in profiles
         self.register_buffer("num_batches_tracked",
                              torch.tensor(0, dtype=torch.long))
 
@@ -124,8 +133,6 @@ class FusedBNReLU(nn.Module):
         if _nhwc_ok(x, self.num_features) and (
                 residual is None or
                 residual.is_contiguous(memory_format=torch.channels_last)):
-            if self.training:
-                self.num_batches_tracked += 1
             res = residual.to(torch.bfloat16) if residual is not None else None
             return _FusedBNFunction.apply(
                 x, res, self.weight, self.bias, self.running_mean,
