@@ -43,7 +43,7 @@ def parse_args():
     p.add_argument("--device", default="cuda" if torch.cuda.is_available()
                    else "cpu")
     p.add_argument("--compression", default="none",
-                   choices=["none", "onebit", "topk", "randomk", "dithering"])
+                   choices=["none", "onebit", "topk", "randomk", "dithering", "fp8"])
     p.add_argument("--partition-mb", type=int, default=0,
                    help="override BPS_PARTITION_BYTES (MiB)")
     return p.parse_args()

@@ -20,7 +20,7 @@ import torch
 from .. import ops as K
 
 # wire codec ids (ops/csrc/kv.h WireCodec)
-RAW, ONEBIT, TOPK, RANDOMK, DITHER_LINEAR, DITHER_NATURAL = range(6)
+RAW, ONEBIT, TOPK, RANDOMK, DITHER_LINEAR, DITHER_NATURAL, FP8 = range(7)
 
 
 class Compressed:
@@ -162,6 +162,22 @@ class DitheringCompressor(BaseCompressor):
         return K.dithering_decompress(code, norm_t, self.s, self.natural, out)
 
 
+class Fp8Compressor(BaseCompressor):
+    """OCP e4m3fn wire with per-partition amax scaling (MI355X-native
+    addition: gfx950's hardware fp8 format; 4× compression, ~2^-3
+    relative precision, deterministic — no RNG)."""
+    codec = FP8
+
+    def compress(self, x: torch.Tensor) -> Compressed:
+        code, amax = K.fp8_compress(x.reshape(-1).float())
+        return Compressed([amax.view(torch.uint8), code])
+
+    def decompress(self, payload, n, aux=0, out=None):
+        amax = payload[:4].view(torch.float32).contiguous()
+        code = payload[4:4 + n].contiguous()
+        return K.fp8_decompress(code, amax, out)
+
+
 class NesterovMomentum(BaseCompressor):
     """m = μm + g ; g += μm before compression (reference
     impl/nesterov_momentum.cc:39-49) — fused HIP kernel on GPU."""
@@ -222,6 +238,7 @@ _REGISTRY = {
         int(p.get("compressor_k", 64)),
         str(p.get("partition", "linear")) == "natural",
         int(p.get("seed", 1))),
+    "fp8": lambda p: Fp8Compressor(),
 }
 
 

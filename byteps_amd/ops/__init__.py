@@ -230,6 +230,42 @@ def sparse_error(x: torch.Tensor, idx: torch.Tensor,
     return err
 
 
+# -- fp8 e4m3 wire -----------------------------------------------------------
+
+def fp8_compress(x: torch.Tensor,
+                 amax_t: Optional[torch.Tensor] = None
+                 ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """→ (codes uint8 [n], amax float[1]).  OCP e4m3fn with per-partition
+    448/amax scaling; GPU and CPU produce identical bytes."""
+    n = x.numel()
+    if amax_t is None:
+        amax_t = norm(x, "max")
+    if x.is_cuda:
+        _check_gpu(x)
+        code = torch.empty(n, dtype=torch.uint8, device=x.device)
+        core().fp8_compress(x.data_ptr(), n, amax_t.data_ptr(),
+                            code.data_ptr(), _stream(x))
+        return code, amax_t
+    code = torch.empty(n, dtype=torch.uint8)
+    core().cpu_fp8_compress(x.float().contiguous().data_ptr(), n,
+                            float(amax_t.item()), code.data_ptr())
+    return code, amax_t
+
+
+def fp8_decompress(code: torch.Tensor, amax_t: torch.Tensor,
+                   out: Optional[torch.Tensor] = None) -> torch.Tensor:
+    n = code.numel()
+    if out is None:
+        out = torch.empty(n, dtype=torch.float32, device=code.device)
+    if code.is_cuda:
+        core().fp8_decompress(code.data_ptr(), n, amax_t.data_ptr(),
+                              out.data_ptr(), _stream(code))
+    else:
+        core().cpu_fp8_decompress(code.data_ptr(), n, float(amax_t.item()),
+                                  out.data_ptr())
+    return out
+
+
 # -- dithering --------------------------------------------------------------
 
 def dithering_compress(x: torch.Tensor, s: int, seed: int,

@@ -82,4 +82,59 @@ BPS_HD inline uint64_t rand_index(uint64_t seed, uint64_t j, uint64_t n) {
   return xorshift128p(s0, s1) % n;
 }
 
+// -- OCP fp8 e4m3fn software conversion --------------------------------------
+// gfx950's native fp8 is OCP e4m3fn (NOT the MI300X fnuz variant).  The
+// software conversion here is shared by the HIP kernels and the CPU
+// server so wire bytes are bit-identical on both sides (exact power-of-2
+// arithmetic + round-half-even everywhere).
+// Layout: s EEEE MMM, bias 7; max normal 0x7E = 448; 0x7F = NaN (unused:
+// we saturate); subnormals step 2^-9.
+
+BPS_HD inline uint8_t fp8_e4m3_encode(float x) {
+  uint8_t sign = x < 0.0f ? 0x80 : 0x00;
+  float a = x < 0.0f ? -x : x;
+  if (!(a > 0.0f)) return sign;                 // ±0 and NaN → 0
+  if (a >= 448.0f) return sign | 0x7E;          // saturate to max normal
+  int e = 0;
+  float m = a;
+  while (m >= 2.0f) { m *= 0.5f; ++e; }
+  while (m < 1.0f && e > -6) { m *= 2.0f; --e; }
+  if (m < 1.0f) {
+    // subnormal: units of 2^-9
+    float q = a * 512.0f;                       // a / 2^-9
+    int qi = (int)(q + 0.5f);
+    if (((float)qi - q == 0.5f) && (qi & 1)) --qi;   // ties to even
+    if (qi <= 0) return sign;
+    if (qi > 7) return sign | 0x08;             // rounds up to 2^-6
+    return sign | (uint8_t)qi;
+  }
+  // normal: mant = round((m-1)*8)
+  float mf = (m - 1.0f) * 8.0f;
+  int mi = (int)(mf + 0.5f);
+  if (((float)mi - mf == 0.5f) && (mi & 1)) --mi;
+  if (mi == 8) { mi = 0; ++e; }
+  if (e > 8) return sign | 0x7E;
+  if (e == 8 && mi > 6) return sign | 0x7E;
+  return sign | (uint8_t)(((e + 7) << 3) | mi);
+}
+
+BPS_HD inline float fp8_e4m3_decode(uint8_t b) {
+  float sign = (b & 0x80) ? -1.0f : 1.0f;
+  int ef = (b >> 3) & 0xF;
+  int mant = b & 0x7;
+  if (ef == 0) {
+    // subnormal: mant * 2^-9
+    return sign * (float)mant * 0.001953125f;
+  }
+  float m = 1.0f + (float)mant * 0.125f;
+  int e = ef - 7;
+  float p = 1.0f;
+  if (e >= 0) {
+    for (int i = 0; i < e; ++i) p *= 2.0f;
+    return sign * m * p;
+  }
+  for (int i = 0; i < -e; ++i) p *= 0.5f;
+  return sign * m * p;
+}
+
 }  // namespace bpsamd

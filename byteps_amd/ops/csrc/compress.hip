@@ -217,6 +217,36 @@ __global__ void dithering_natural_decompress_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// fp8 e4m3fn wire (MI355X-native addition — no reference counterpart):
+// 4× compression with ~2^-3 relative precision.  Values are scaled by
+// 448/amax before encoding so the full fp8 range is used; the amax rides
+// in front of the codes (wire: [f32 amax][u8 codes]).
+// ---------------------------------------------------------------------------
+
+__global__ void fp8_compress_kernel(const float* __restrict__ x, int64_t n,
+                                    const float* __restrict__ amax,
+                                    uint8_t* __restrict__ code) {
+  const float a = amax[0];
+  const float scale = a > 0.0f ? 448.0f / a : 0.0f;
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride)
+    code[i] = bpsamd::fp8_e4m3_encode(x[i] * scale);
+}
+
+__global__ void fp8_decompress_kernel(const uint8_t* __restrict__ code,
+                                      int64_t n,
+                                      const float* __restrict__ amax,
+                                      float* __restrict__ out) {
+  const float a = amax[0];
+  const float inv = a / 448.0f;
+  int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (int64_t i = i0; i < n; i += stride)
+    out[i] = bpsamd::fp8_e4m3_decode(code[i]) * inv;
+}
+
 }  // namespace
 
 #define STREAM reinterpret_cast<hipStream_t>(stream)
@@ -303,6 +333,22 @@ int bps_dithering_decompress(const void* code, int64_t n, int s, int natural,
     hipLaunchKernelGGL(dithering_linear_decompress_kernel, dim3(grid_for(n)),
                        dim3(BLOCK), 0, STREAM, (const int8_t*)code, n, s,
                        (const float*)norm, (float*)out);
+  return (int)hipGetLastError();
+}
+
+int bps_fp8_compress(const void* x, int64_t n, const void* amax, void* code,
+                     void* stream) {
+  hipLaunchKernelGGL(fp8_compress_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
+                     STREAM, (const float*)x, n, (const float*)amax,
+                     (uint8_t*)code);
+  return (int)hipGetLastError();
+}
+
+int bps_fp8_decompress(const void* code, int64_t n, const void* amax,
+                       void* out, void* stream) {
+  hipLaunchKernelGGL(fp8_decompress_kernel, dim3(grid_for(n)), dim3(BLOCK),
+                     0, STREAM, (const uint8_t*)code, n, (const float*)amax,
+                     (float*)out);
   return (int)hipGetLastError();
 }
 

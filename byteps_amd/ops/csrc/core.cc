@@ -43,6 +43,10 @@ int bps_dithering_compress(const void* x, int64_t n, int s, uint64_t seed,
                            void* stream);
 int bps_dithering_decompress(const void* code, int64_t n, int s, int natural,
                              const void* norm, void* out, void* stream);
+int bps_fp8_compress(const void* x, int64_t n, const void* amax, void* code,
+                     void* stream);
+int bps_fp8_decompress(const void* code, int64_t n, const void* amax,
+                       void* out, void* stream);
 
 // -- fused batchnorm (bn.hip) ----------------------------------------------
 int bps_bn_reduce(const void* x, long long M, int C, void* sums,
@@ -97,6 +101,10 @@ int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
                                int natural, float norm, int8_t* code);
 int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
                                  int natural, float norm, float* out);
+int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
+                         uint8_t* code);
+int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
+                           float* out);
 float bps_cpu_norm(const float* x, int64_t n, int mode);
 }
 
@@ -190,6 +198,31 @@ PYBIND11_MODULE(_core, m) {
           check(bps_dithering_decompress(CP(code), n, sv, natural, CP(norm),
                                          P(out), P(s)),
                 "bps_dithering_decompress");
+        });
+
+  m.def("fp8_compress",
+        [](uintptr_t x, int64_t n, uintptr_t amax, uintptr_t code,
+           uintptr_t s) {
+          check(bps_fp8_compress(CP(x), n, CP(amax), P(code), P(s)),
+                "bps_fp8_compress");
+        });
+  m.def("fp8_decompress",
+        [](uintptr_t code, int64_t n, uintptr_t amax, uintptr_t out,
+           uintptr_t s) {
+          check(bps_fp8_decompress(CP(code), n, CP(amax), P(out), P(s)),
+                "bps_fp8_decompress");
+        });
+  m.def("cpu_fp8_compress",
+        [](uintptr_t x, int64_t n, float amax, uintptr_t code) {
+          check(bps_cpu_fp8_compress((const float*)P(x), n, amax,
+                                     (uint8_t*)P(code)),
+                "bps_cpu_fp8_compress");
+        });
+  m.def("cpu_fp8_decompress",
+        [](uintptr_t code, int64_t n, float amax, uintptr_t out) {
+          check(bps_cpu_fp8_decompress((const uint8_t*)P(code), n, amax,
+                                       (float*)P(out)),
+                "bps_cpu_fp8_decompress");
         });
 
   // fused batchnorm

@@ -208,6 +208,24 @@ int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
   return 0;
 }
 
+int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
+                         uint8_t* code) {
+  const float scale = amax > 0.0f ? 448.0f / amax : 0.0f;
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i)
+    code[i] = fp8_e4m3_encode(x[i] * scale);
+  return 0;
+}
+
+int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
+                           float* out) {
+  const float inv = amax / 448.0f;
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i)
+    out[i] = fp8_e4m3_decode(code[i]) * inv;
+  return 0;
+}
+
 float bps_cpu_norm(const float* x, int64_t n, int mode) {
   if (mode == 2) {
     float mx = 0.0f;

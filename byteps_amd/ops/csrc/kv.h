@@ -31,6 +31,7 @@ enum WireCodec : uint32_t {
   kRandomk = 3,
   kDitherLinear = 4,
   kDitherNatural = 5,
+  kFp8 = 6,            // OCP e4m3fn with per-partition amax scale
 };
 
 // cmd encoding: low 8 bits codec, next 8 bits dtype, bit 16 async-mode

@@ -66,6 +66,10 @@ int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
                                int natural, float norm, int8_t* code);
 int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
                                  int natural, float norm, float* out);
+int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
+                         uint8_t* code);
+int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
+                           float* out);
 float bps_cpu_norm(const float* x, int64_t n, int mode);
 }
 
@@ -487,6 +491,18 @@ class Server {
           bps_cpu_sum(acc, ks->scratch.data(), n, 0);
         break;
       }
+      case kFp8: {
+        float amax;
+        std::memcpy(&amax, t.payload.data(), 4);
+        const uint8_t* code = (const uint8_t*)(t.payload.data() + 4);
+        ks->scratch.resize(n);
+        bps_cpu_fp8_decompress(code, n, amax, ks->scratch.data());
+        if (first)
+          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
+        else
+          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        break;
+      }
       default:
         break;
     }
@@ -589,6 +605,14 @@ class Server {
                                    norm, (int8_t*)(ks->reply.data() + 4));
         break;
       }
+      case kFp8: {
+        float amax = bps_cpu_norm(acc, n, 2);
+        ks->reply.resize(4 + n);
+        std::memcpy(ks->reply.data(), &amax, 4);
+        bps_cpu_fp8_compress(acc, n, amax,
+                             (uint8_t*)(ks->reply.data() + 4));
+        break;
+      }
       default:
         break;
     }
@@ -621,6 +645,13 @@ class Server {
           bps_cpu_dithering_decompress(
               (const int8_t*)(ks->reply.data() + 4), n, (int)ks->levels,
               ks->codec == kDitherNatural, norm, dec);
+          break;
+        }
+        case kFp8: {
+          float amax;
+          std::memcpy(&amax, ks->reply.data(), 4);
+          bps_cpu_fp8_decompress((const uint8_t*)(ks->reply.data() + 4), n,
+                                 amax, dec);
           break;
         }
         default:

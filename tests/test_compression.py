@@ -139,3 +139,25 @@ def test_nesterov_momentum_math():
     m2.mul_(0.9).add_(g2)
     g2.add_(m2, alpha=0.9)
     assert torch.allclose(g, g2) and torch.allclose(m, m2)
+
+
+def test_fp8_roundtrip_and_bounds():
+    x = _rand(10000, seed=6) * 3
+    c = comp.Fp8Compressor()
+    cp = c.compress(x)
+    assert cp.nbytes == 10000 + 4     # 4x compression + amax
+    out = c.decompress(comp.BaseCompressor._payload_cat(cp), 10000)
+    amax = x.abs().max()
+    rel = (out - x).abs() / (x.abs() + 1e-9)
+    normal = x.abs() > float(amax) / 448 * 2 ** -6
+    assert rel[normal].max() < 1 / 16 + 1e-3          # e4m3 mantissa bound
+    assert (out - x).abs()[~normal].max() <= float(amax) / 448 * 2 ** -6
+    # deterministic (no RNG)
+    cp2 = comp.Fp8Compressor().compress(x)
+    assert torch.equal(cp.parts[1], cp2.parts[1])
+
+
+def test_fp8_registry():
+    c = comp.create({"compressor_type": "fp8", "ef_type": "vanilla"})
+    assert isinstance(c, comp.ErrorFeedback)
+    assert c.codec == comp.FP8

@@ -157,3 +157,19 @@ def test_compressor_classes_on_gpu():
         out = c.decompress(payload, n, cp.aux)
         assert out.shape[0] == n and out.is_cuda
         torch.cuda.synchronize()
+
+
+def test_fp8_gpu_matches_cpu():
+    n = 100000
+    xg = _cuda(n, 13) * 5
+    code_g, amax_g = K.fp8_compress(xg)
+    torch.cuda.synchronize()
+    code_c, amax_c = K.fp8_compress(xg.cpu(), amax_t=amax_g.cpu())
+    assert torch.equal(code_g.cpu(), code_c), "fp8 bytes must be bit-identical"
+    out = K.fp8_decompress(code_g, amax_g)
+    torch.cuda.synchronize()
+    out_c = K.fp8_decompress(code_c, amax_c)
+    assert torch.allclose(out.cpu(), out_c)
+    rel = (out - xg).abs() / (xg.abs() + 1e-9)
+    normal = xg.abs() > amax_g / 448 * 2 ** -6
+    assert rel[normal].max().item() < 1 / 16 + 1e-3

@@ -248,3 +248,37 @@ def test_ps_three_ranks_one_node_parity(server):
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def _ps_fp8(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    opt = bps.DistributedOptimizer(
+        opt, named_parameters=m.named_parameters(),
+        compression_params={"compressor_type": "fp8", "ef_type": "vanilla"})
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = ((m(xs[rank]) - ys[rank]) ** 2).mean()
+        loss.backward()
+        losses.append(float(loss))
+        opt.step()
+    bps.shutdown()
+    return losses
+
+
+def test_ps_fp8_trains(server):
+    """fp8 e4m3 wire through the full worker↔server loop: near-lossless
+    (2^-3 relative), loss must track the uncompressed trajectory."""
+    results = run_in_processes(
+        _ps_fp8, 2, 20,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1",
+                                   "BPS_MIN_COMPRESS_BYTES": "0"}))
+    for losses in results:
+        assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
