@@ -259,12 +259,14 @@ def _ps_fp8(rank, world, steps):
         opt, named_parameters=m.named_parameters(),
         compression_params={"compressor_type": "fp8", "ef_type": "vanilla"})
     torch.manual_seed(42)
-    xs = [torch.randn(8, 16) for _ in range(world)]
-    ys = [torch.randn(8, 4) for _ in range(world)]
+    # same regressable data on every rank → loss floor near 0
+    x = torch.randn(16, 16)
+    w = torch.randn(16, 4)
+    y = x @ w * 0.3
     losses = []
     for _ in range(steps):
         opt.zero_grad()
-        loss = ((m(xs[rank]) - ys[rank]) ** 2).mean()
+        loss = ((m(x) - y) ** 2).mean()
         loss.backward()
         losses.append(float(loss))
         opt.step()
