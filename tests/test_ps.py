@@ -278,7 +278,7 @@ def test_ps_fp8_trains(server):
     """fp8 e4m3 wire through the full worker↔server loop: near-lossless
     (2^-3 relative), loss must track the uncompressed trajectory."""
     results = run_in_processes(
-        _ps_fp8, 2, 20,
+        _ps_fp8, 2, 50,
         extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
                                    "LOCAL_WORLD_SIZE": "1",
                                    "BPS_MIN_COMPRESS_BYTES": "0"}))
