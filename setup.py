@@ -32,7 +32,8 @@ setup(
     description="MI355X-native gradient-synchronization framework "
                 "(BytePS-capability, HIP/CDNA4 + RCCL + native PS)",
     packages=find_packages(include=["byteps_amd", "byteps_amd.*"]),
-    package_data={"byteps_amd.ops": ["_core.so", "csrc/*"]},
+    package_data={"byteps_amd.ops": ["_core.so", "csrc/*"],
+                  "byteps_amd": ["tuning/*.csv"]},
     python_requires=">=3.8",
     cmdclass={"build_py": BuildWithNative},
     entry_points={
