@@ -90,51 +90,48 @@ BPS_HD inline uint64_t rand_index(uint64_t seed, uint64_t j, uint64_t n) {
 // Layout: s EEEE MMM, bias 7; max normal 0x7E = 448; 0x7F = NaN (unused:
 // we saturate); subnormals step 2^-9.
 
+BPS_HD inline uint32_t f32_bits(float f) {
+  uint32_t u;
+  __builtin_memcpy(&u, &f, 4);
+  return u;
+}
+
+BPS_HD inline float bits_f32(uint32_t u) {
+  float f;
+  __builtin_memcpy(&f, &u, 4);
+  return f;
+}
+
+// Branch-light bit-manipulation conversion — integer ops only on the
+// normal path, so host and device produce identical bytes.
 BPS_HD inline uint8_t fp8_e4m3_encode(float x) {
-  uint8_t sign = x < 0.0f ? 0x80 : 0x00;
-  float a = x < 0.0f ? -x : x;
-  if (!(a > 0.0f)) return sign;                 // ±0 and NaN → 0
-  if (a >= 448.0f) return sign | 0x7E;          // saturate to max normal
-  int e = 0;
-  float m = a;
-  while (m >= 2.0f) { m *= 0.5f; ++e; }
-  while (m < 1.0f && e > -6) { m *= 2.0f; --e; }
-  if (m < 1.0f) {
-    // subnormal: units of 2^-9
-    float q = a * 512.0f;                       // a / 2^-9
+  uint32_t u = f32_bits(x);
+  uint8_t sign = (uint8_t)((u >> 31) << 7);
+  uint32_t au = u & 0x7FFFFFFFu;
+  if (au > 0x7F800000u) return sign;            // NaN → 0
+  if (au >= 0x43E00000u) return sign | 0x7E;    // |x| ≥ 448 → max normal
+  if (au < 0x3C800000u) {                       // |x| < 2^-6 → subnormal
+    float q = bits_f32(au) * 512.0f;            // units of 2^-9, q ∈ [0, 32)
     int qi = (int)(q + 0.5f);
     if (((float)qi - q == 0.5f) && (qi & 1)) --qi;   // ties to even
-    if (qi <= 0) return sign;
-    if (qi > 7) return sign | 0x08;             // rounds up to 2^-6
-    return sign | (uint8_t)qi;
+    return sign | (uint8_t)qi;                  // qi == 8 encodes 2^-6
   }
-  // normal: mant = round((m-1)*8)
-  float mf = (m - 1.0f) * 8.0f;
-  int mi = (int)(mf + 0.5f);
-  if (((float)mi - mf == 0.5f) && (mi & 1)) --mi;
-  if (mi == 8) { mi = 0; ++e; }
-  if (e > 8) return sign | 0x7E;
-  if (e == 8 && mi > 6) return sign | 0x7E;
-  return sign | (uint8_t)(((e + 7) << 3) | mi);
+  // normal: RNE-round the f32 mantissa down to 3 bits via integer add
+  uint32_t lsb = (au >> 20) & 1u;
+  au += 0x0007FFFFu + lsb;
+  int ef = (int)(au >> 23) - 127 + 7;
+  uint32_t mant = (au >> 20) & 7u;
+  if (ef > 15 || (ef == 15 && mant > 6)) return sign | 0x7E;
+  return sign | (uint8_t)((ef << 3) | mant);
 }
 
 BPS_HD inline float fp8_e4m3_decode(uint8_t b) {
-  float sign = (b & 0x80) ? -1.0f : 1.0f;
-  int ef = (b >> 3) & 0xF;
-  int mant = b & 0x7;
-  if (ef == 0) {
-    // subnormal: mant * 2^-9
-    return sign * (float)mant * 0.001953125f;
-  }
-  float m = 1.0f + (float)mant * 0.125f;
-  int e = ef - 7;
-  float p = 1.0f;
-  if (e >= 0) {
-    for (int i = 0; i < e; ++i) p *= 2.0f;
-    return sign * m * p;
-  }
-  for (int i = 0; i < -e; ++i) p *= 0.5f;
-  return sign * m * p;
+  const int ef = (b >> 3) & 0xF;
+  const uint32_t mant = b & 0x7u;
+  const uint32_t s = ((uint32_t)(b >> 7)) << 31;
+  if (ef == 0)                                  // subnormal: mant × 2^-9
+    return bits_f32(s | 0x3B000000u /*2^-9*/) * (float)mant;
+  return bits_f32(s | ((uint32_t)(ef - 7 + 127) << 23) | (mant << 20));
 }
 
 }  // namespace bpsamd
