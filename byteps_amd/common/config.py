@@ -9,7 +9,7 @@ working.  Every knob is read lazily so tests can monkeypatch ``os.environ``.
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional
 
 _TRUE = {"1", "true", "yes", "on"}
@@ -92,8 +92,6 @@ class Config:
 
     # reduction -------------------------------------------------------------
     reduce_dtype: str = ""                  # "" → same as grad; "fp32" to upcast
-
-    extra: dict = field(default_factory=dict)
 
     @staticmethod
     def from_env() -> "Config":

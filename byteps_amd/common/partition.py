@@ -71,7 +71,6 @@ def plan_partitions(
 
     for pidx, size in enumerate(sizes):
         remaining = size
-        consumed = 0
         while remaining > 0:
             space = partition_elems - cur_off
             if space < align:
@@ -81,7 +80,6 @@ def plan_partitions(
             cur_spans.append(Span(pidx, cur_off, take))
             cur_off = _align_up(cur_off + take, align)
             remaining -= take
-            consumed += take
     flush()
 
     n = len(plans)

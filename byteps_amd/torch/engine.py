@@ -75,7 +75,6 @@ class Bucket:
     issued: bool = False
     work: Optional[object] = None             # dist Work handle
     ps_ticket: Optional[object] = None        # PS pipeline ticket
-    done_event: Optional[torch.cuda.Event] = None
 
     @property
     def nbytes(self) -> int:
@@ -86,7 +85,6 @@ class Bucket:
         self.issued = False
         self.work = None
         self.ps_ticket = None
-        self.done_event = None
 
 
 # --------------------------------------------------------------------------

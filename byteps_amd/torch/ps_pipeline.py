@@ -75,7 +75,6 @@ class _Staging:
         self.recv = torch.empty(nbytes_recv, dtype=torch.uint8,
                                 pin_memory=pin)
         self.stream = torch.cuda.Stream(device) if self.on_gpu else None
-        self.gpu_scratch = None
 
 
 class Ticket:
