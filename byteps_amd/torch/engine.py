@@ -29,7 +29,7 @@ from __future__ import annotations
 
 import heapq
 import threading
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Dict, List, Optional, Sequence
 
 import torch
@@ -121,7 +121,6 @@ class GradEngine:
         self._pending: List = []              # heap of (-priority, idx)
         self._inflight_bytes = 0
         self._credit = self.cfg.scheduling_credit  # 0 → unlimited
-        self._deferred: List[int] = []
         self._hook_handles: List = []
         self._sync_enabled = True
         self._step = 0
@@ -165,7 +164,6 @@ class GradEngine:
 
         self.buckets: List[Bucket] = []
         self.param_bucket: Dict[int, List[Bucket]] = {}   # param idx → buckets
-        self._param_span_count: Dict[int, int] = {i: 0 for i in range(len(self.params))}
         for plan in plans:
             buf = torch.zeros(plan.numel, dtype=dt, device=dev)
             bucket = Bucket(plan=plan, buffer=buf, params=[], grads=[],
@@ -180,7 +178,6 @@ class GradEngine:
                 bucket.params.append(p)
                 bucket.grads.append(view)
                 self.param_bucket.setdefault(pidx, []).append(bucket)
-                self._param_span_count[pidx] += 1
             self.buckets.append(bucket)
 
         # Attach p.grad views.  A param split across buckets cannot be a
@@ -225,7 +222,6 @@ class GradEngine:
             self._ps = ps_pipeline.get_pipeline(self)
 
         self._attach_hooks()
-        self._expected_ready = {id(b): len(b.params) for b in self.buckets}
         log.debug("GradEngine: %d params → %d buckets (%.1f MiB each max)",
                   len(self.params), len(self.buckets),
                   part_elems * buf.element_size() / 2**20)
