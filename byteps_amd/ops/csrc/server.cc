@@ -218,8 +218,11 @@ class Server {
     for (auto& w : workers_) {
       if (w.joinable()) w.join();
     }
-    for (auto& r : readers_) {
-      if (r.joinable()) r.join();
+    {
+      std::lock_guard<std::mutex> lk(readers_mu_);
+      for (auto& r : readers_) {
+        if (r.joinable()) r.join();
+      }
     }
     {
       std::lock_guard<std::mutex> lk(conns_mu_);
@@ -284,7 +287,10 @@ class Server {
         std::lock_guard<std::mutex> lk(conns_mu_);
         conns_.push_back(conn);
       }
-      readers_.emplace_back([this, conn] { reader_loop(conn); });
+      {
+        std::lock_guard<std::mutex> lk(readers_mu_);
+        readers_.emplace_back([this, conn] { reader_loop(conn); });
+      }
     }
   }
 
@@ -684,6 +690,7 @@ class Server {
   bool enable_schedule_;
   std::atomic<bool> running_{false};
   std::thread accept_thread_;
+  std::mutex readers_mu_;
   std::vector<std::thread> readers_;
   std::vector<std::thread> workers_;
   std::vector<std::unique_ptr<EngineQueue>> queues_;
