@@ -65,18 +65,37 @@ def _sdpa_ctx(seq_len: int):
 
 
 class SelfAttention(nn.Module):
+    """Packed qkv (one GEMM) by default.  ``BPS_BERT_SPLIT_QKV=1`` uses
+    three separate projections: slightly smaller GEMMs, but no per-layer
+    unbind→contiguous copies forward and no CatArrayBatchedCopy grad
+    assembly backward (profiles/bert_large_steady_state.md showed those
+    at ~1.5 ms/step)."""
+
     def __init__(self, cfg: BertConfig):
         super().__init__()
+        import os
         self.heads = cfg.heads
         self.head_dim = cfg.hidden // cfg.heads
-        self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
+        self.split_qkv = os.environ.get("BPS_BERT_SPLIT_QKV", "0") == "1"
+        if self.split_qkv:
+            self.q_proj = nn.Linear(cfg.hidden, cfg.hidden)
+            self.k_proj = nn.Linear(cfg.hidden, cfg.hidden)
+            self.v_proj = nn.Linear(cfg.hidden, cfg.hidden)
+        else:
+            self.qkv = nn.Linear(cfg.hidden, 3 * cfg.hidden)
         self.out = nn.Linear(cfg.hidden, cfg.hidden)
         self.dropout = cfg.dropout
 
     def forward(self, x):
         B, S, H = x.shape
-        qkv = self.qkv(x).view(B, S, 3, self.heads, self.head_dim)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4)          # 3 × (B, h, S, d)
+        h, d = self.heads, self.head_dim
+        if self.split_qkv:
+            q = self.q_proj(x).view(B, S, h, d).transpose(1, 2)
+            k = self.k_proj(x).view(B, S, h, d).transpose(1, 2)
+            v = self.v_proj(x).view(B, S, h, d).transpose(1, 2)
+        else:
+            qkv = self.qkv(x).view(B, S, 3, h, d)
+            q, k, v = qkv.permute(2, 0, 3, 1, 4)      # 3 × (B, h, S, d)
         ctx = _sdpa_ctx(S) if x.is_cuda else None
         if ctx is not None:
             with ctx:
