@@ -284,3 +284,17 @@ def test_ps_fp8_trains(server):
                                    "BPS_MIN_COMPRESS_BYTES": "0"}))
     for losses in results:
         assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
+
+
+def test_ps_two_nodes_two_ranks_each(server):
+    """4 ranks as 2 simulated nodes × 2 GPUs: intra-node reduce within
+    node subgroups, per-shard push with expected pushers = 2 nodes,
+    all-gather within each node — the full hierarchical topology."""
+    expected = _baseline(4, 3)
+    results = run_in_processes(
+        _ps_worker, 4, 3, None,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "2",
+                                   "LOCAL_WORLD_SIZE": "2"}))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
