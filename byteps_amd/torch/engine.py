@@ -70,6 +70,7 @@ class Bucket:
     grads: List[torch.Tensor]                 # the views (same order)
     priority: int = 0
     declared_key: int = 0
+    compression_params: Optional[dict] = None   # resolved codec config
     # per-step state
     ready_count: int = 0
     issued: bool = False
@@ -148,37 +149,68 @@ class GradEngine:
         part_bytes = partition_bytes or self.cfg.partition_bytes
         part_elems = max(part_bytes // torch.tensor([], dtype=dt).element_size(), 4096)
 
+        # Per-parameter compression overrides (reference: per-param
+        # byteps_* attrs, mxnet/__init__.py:250-317): params whose
+        # resolved codec config differs go into separate bucket groups so
+        # each bucket has ONE wire format.  "param_overrides" maps a name
+        # substring → kwargs merged over the base config.
+        overrides = self.compression_params.get("param_overrides") or {}
+        base_cfg = {k: v for k, v in self.compression_params.items()
+                    if k != "param_overrides"}
+
+        def resolved_cfg(name: str) -> dict:
+            cfg = dict(base_cfg)
+            for pat, extra in overrides.items():
+                if pat in name:
+                    cfg.update(extra)
+            return cfg
+
         # Bucket in REVERSE registration order: backward runs output → input,
         # so the last-registered params produce gradients first and land in
         # bucket 0 (highest priority) — the priority semantics of the
         # reference (mxnet/__init__.py:58-60, scheduled_queue.cc:86-95).
         rev = list(range(len(self.params)))[::-1]
-        sizes = [self.params[i].numel() for i in rev]
+        groups: Dict[str, List[int]] = {}
+        group_cfg: Dict[str, dict] = {}
+        for i in rev:
+            cfg = resolved_cfg(self.param_names[i])
+            gkey = repr(sorted(cfg.items()))
+            groups.setdefault(gkey, []).append(i)
+            group_cfg[gkey] = cfg
         # align so reduce-scatter shards divide evenly for THIS world:
         # lcm(64, world) keeps 16-byte vector alignment for the HIP
         # kernels and exact shard division for any rank count (a plain 64
         # silently truncates the bucket tail at e.g. world=3 or 6)
         import math
         align = 64 * self.world // math.gcd(64, self.world)
-        plans = plan_partitions(sizes, part_elems, align=align)
 
         self.buckets: List[Bucket] = []
         self.param_bucket: Dict[int, List[Bucket]] = {}   # param idx → buckets
-        for plan in plans:
-            buf = torch.zeros(plan.numel, dtype=dt, device=dev)
-            bucket = Bucket(plan=plan, buffer=buf, params=[], grads=[],
-                            priority=plan.priority)
-            key = C._state.registry.declare(
-                "byteps.Partition.%d" % plan.index)
-            bucket.declared_key = key
-            for span in plan.spans:
-                pidx = rev[span.param_index]
-                p = self.params[pidx]
-                view = buf.narrow(0, span.offset, span.numel)
-                bucket.params.append(p)
-                bucket.grads.append(view)
-                self.param_bucket.setdefault(pidx, []).append(bucket)
-            self.buckets.append(bucket)
+        for gkey, idxs in groups.items():
+            sizes = [self.params[i].numel() for i in idxs]
+            plans = plan_partitions(sizes, part_elems, align=align)
+            for plan in plans:
+                buf = torch.zeros(plan.numel, dtype=dt, device=dev)
+                index = len(self.buckets)
+                bucket = Bucket(plan=plan, buffer=buf, params=[], grads=[])
+                bucket.plan.index = index
+                key = C._state.registry.declare(
+                    "byteps.Partition.%d" % index)
+                bucket.declared_key = key
+                bucket.compression_params = group_cfg[gkey]
+                for span in plan.spans:
+                    pidx = idxs[span.param_index]
+                    p = self.params[pidx]
+                    view = buf.narrow(0, span.offset, span.numel)
+                    bucket.params.append(p)
+                    bucket.grads.append(view)
+                    self.param_bucket.setdefault(pidx, []).append(bucket)
+                self.buckets.append(bucket)
+        # priorities follow creation order (reverse-registration within
+        # each group, groups in first-seen order)
+        n = len(self.buckets)
+        for i, b in enumerate(self.buckets):
+            b.priority = n - i
 
         # Attach p.grad views.  A param split across buckets cannot be a
         # single view — those (rare: only params > partition size) get a

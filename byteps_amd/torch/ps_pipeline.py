@@ -52,16 +52,18 @@ def _make_cmd(codec: int, dtype: int, async_mode: bool) -> int:
 
 class _KeyInfo:
     __slots__ = ("key", "server", "nelem", "round", "initialized",
-                 "compressor")
+                 "compressor", "server_ef")
 
     def __init__(self, key: int, server: int, nelem: int,
-                 compressor: Optional[BaseCompressor]):
+                 compressor: Optional[BaseCompressor],
+                 server_ef: bool = False):
         self.key = key
         self.server = server
         self.nelem = nelem
         self.round = 0
         self.initialized = False
         self.compressor = compressor
+        self.server_ef = server_ef
 
 
 class _Staging:
@@ -154,9 +156,8 @@ class PSPipeline:
         self.node_world = dist.get_world_size(self.node_group) \
             if (self.world > 1 and dist.is_initialized()) else 1
 
-        # per-tensor compressor instances (engine-wide config for now;
-        # per-parameter params like the reference's byteps_* attrs can be
-        # layered on via set_compression_params)
+        # engine-wide defaults; each bucket carries its own resolved
+        # config (per-parameter overrides, reference byteps_* attrs)
         self.compression_params: Dict = dict(
             getattr(engine, "compression_params", {}) or {})
 
@@ -176,10 +177,17 @@ class PSPipeline:
                 nbytes = nelem * 4
                 server = C._state.assigner.assign(pkey, nbytes)
                 compressor = None
-                if self.compression_params.get("compressor_type") and \
+                cfg = dict(bucket.compression_params
+                           if bucket.compression_params is not None
+                           else self.compression_params)
+                cfg.pop("param_overrides", None)
+                if cfg.get("compressor_type") and \
                         nbytes >= self.cfg.min_compress_bytes:
-                    compressor = create(dict(self.compression_params))
-                ki = _KeyInfo(pkey, server, nelem, compressor)
+                    compressor = create(cfg)
+                ef = str(cfg.get("ef_type", "")).lower() in \
+                    ("vanilla", "1", "true")
+                ki = _KeyInfo(pkey, server, nelem, compressor,
+                              compressor is not None and ef)
                 self.keys[bucket.plan.index] = ki
             return ki
 
@@ -210,9 +218,7 @@ class PSPipeline:
         codec = comp.codec if comp is not None else 0
         levels = comp.levels if comp is not None else 0
         # bit0: ask the server to run error feedback on its merged reply
-        flags = 1 if (comp is not None and str(
-            self.compression_params.get("ef_type", "")).lower()
-            in ("vanilla", "1", "true")) else 0
+        flags = 1 if ki.server_ef else 0
         payload = struct.pack("<QIIII", ki.nelem, self.num_nodes, levels,
                               flags, 0)
         buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)

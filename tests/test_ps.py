@@ -298,3 +298,41 @@ def test_ps_two_nodes_two_ranks_each(server):
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def _ps_overrides(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    # lossless topk for layer "0.", raw for everything else — exact parity
+    opt = bps.DistributedOptimizer(
+        opt, named_parameters=m.named_parameters(),
+        compression_params={
+            "param_overrides": {"0.": {"compressor_type": "topk",
+                                       "compressor_k": 1 << 20}}})
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_ps_per_param_compression_overrides(server):
+    """param_overrides route different params through different wire
+    codecs (reference per-param byteps_* attrs) — lossless settings must
+    preserve exact parity."""
+    expected = _baseline(2, 2)
+    results = run_in_processes(
+        _ps_overrides, 2, 2,
+        extra_env=_ps_env(server, {"BPS_LOCAL_SIZE": "1",
+                                   "LOCAL_WORLD_SIZE": "1",
+                                   "BPS_MIN_COMPRESS_BYTES": "0"}))
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-4, atol=1e-5)
