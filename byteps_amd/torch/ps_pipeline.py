@@ -321,6 +321,13 @@ class PSPipeline:
                                 ki.round)
         reply_len, _ver = self.kv.wait(t_pull)
         _tr("pull", False)
+        if reply_len > st.recv.numel():
+            # the transport drops oversized payloads into scratch — the
+            # staging buffer would hold stale bytes; fail loudly
+            raise RuntimeError(
+                "PS pull reply (%d B) exceeds staging capacity (%d B) for "
+                "key %d — codec config mismatch between worker and server"
+                % (reply_len, st.recv.numel(), ki.key))
         telemetry.record(reply_len)
 
         _tr("h2d+decompress", True)
