@@ -34,7 +34,7 @@ __all__ = [
     "push_pull_group_sync_inplace", "poll", "synchronize", "declare",
     "set_num_grads", "DistributedOptimizer", "broadcast_parameters",
     "broadcast_optimizer_state", "broadcast_object", "Compression",
-    "BytePSPushPull",
+    "BytePSPushPull", "metric_average",
 ]
 
 
@@ -149,6 +149,16 @@ def DistributedOptimizer(optimizer, named_parameters=None,
 # Broadcasts
 # --------------------------------------------------------------------------
 
+def metric_average(value, name: str):
+    """Average a scalar metric across all ranks (reference keras
+    MetricAverageCallback, _keras/callbacks.py:62-86)."""
+    _C._require_init()
+    t = value.detach().clone().float() if torch.is_tensor(value) \
+        else torch.tensor(float(value))
+    avg = push_pull(t, average=True, name="byteps.Metric." + name)
+    return avg.item() if avg.numel() == 1 else avg
+
+
 def broadcast_parameters(params, root_rank: int = 0) -> None:
     """Broadcast a ``state_dict()`` or list of (name, tensor) pairs from
     ``root_rank`` (reference torch/__init__.py:268-299 — which emulated
@@ -209,16 +219,31 @@ def broadcast_optimizer_state(optimizer, root_rank: int = 0) -> None:
         raise ValueError("cannot broadcast torch.optim.LBFGS state")
 
     state_dict = optimizer.state_dict()
-    # materialize state on non-root ranks by running a dummy step if empty
-    # (reference creates state lazily the same way, torch/__init__.py:318-335)
-    if _C.rank() != root_rank and not state_dict["state"]:
+    # Lazy optimizer state must exist CONSISTENTLY on every rank before
+    # the tensor broadcast (mismatched per-rank tensor lists deadlock).
+    # The decision is driven by the ROOT's state: if the root has none
+    # (fresh optimizer), every rank materializes it with a dummy step
+    # (reference's lazy-creation idea, torch/__init__.py:318-335, made
+    # symmetric).  The dummy step bypasses our synchronizing wrapper and
+    # restores params afterwards (weight decay would perturb them).
+    root_has_state = broadcast_object(
+        bool(state_dict["state"]), root_rank, name="byteps.opt_has_state")
+    if not root_has_state or not state_dict["state"]:
         for group in optimizer.param_groups:
             for p in group["params"]:
                 if p.requires_grad and p.grad is None:
                     p.grad = torch.zeros_like(p)
-        had = [bool(g["params"]) for g in optimizer.param_groups]
-        if any(had):
-            optimizer.step()
+        if any(g["params"] for g in optimizer.param_groups):
+            snap = [p.detach().clone() for g in optimizer.param_groups
+                    for p in g["params"]]
+            if hasattr(optimizer, "_engine"):
+                type(optimizer).__mro__[1].step(optimizer)
+            else:
+                optimizer.step()
+            with torch.no_grad():
+                for p, s in zip((p for g in optimizer.param_groups
+                                 for p in g["params"]), snap):
+                    p.copy_(s)
             state_dict = optimizer.state_dict()
 
     scalars = collections.OrderedDict()
