@@ -37,7 +37,7 @@ def _entry(rank: int, world: int, port: int, fn, args, q, extra_env):
         raise
 
 
-def run_in_processes(fn, world: int, *args, extra_env=None, timeout: float = 120.0):
+def run_in_processes(fn, world: int, *args, extra_env=None, timeout: float = 300.0):
     """Spawn ``world`` processes running ``fn(rank, world, *args)`` with a
     gloo-compatible rendezvous on 127.0.0.1.  Returns [result_rank0, ...].
     Raises on any rank failure."""
