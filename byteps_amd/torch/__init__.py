@@ -21,6 +21,7 @@ import torch.distributed as dist
 from .. import common as _C
 from ..common import (init, shutdown, suspend, resume, rank, size,
                       local_rank, local_size, initialized)
+from ..common.telemetry import get_pushpull_speed
 from .compression import Compression
 from .engine import GradEngine, register_engine
 from .ops import (push_pull, push_pull_async, push_pull_inplace,
@@ -34,7 +35,7 @@ __all__ = [
     "push_pull_group_sync_inplace", "poll", "synchronize", "declare",
     "set_num_grads", "DistributedOptimizer", "broadcast_parameters",
     "broadcast_optimizer_state", "broadcast_object", "Compression",
-    "BytePSPushPull", "metric_average",
+    "BytePSPushPull", "metric_average", "get_pushpull_speed",
 ]
 
 
