@@ -13,7 +13,7 @@ state), created from a kwargs dict like the reference's
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 

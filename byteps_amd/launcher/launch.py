@@ -22,7 +22,6 @@ import os
 import signal
 import subprocess
 import sys
-import threading
 from typing import List
 
 

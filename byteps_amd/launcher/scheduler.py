@@ -17,7 +17,7 @@ import os
 import socket
 import socketserver
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from ..common.logging_util import get_logger
 

@@ -8,7 +8,6 @@ falling back to eager PyTorch.
 
 from __future__ import annotations
 
-import os
 from typing import Optional, Tuple
 
 import torch

@@ -13,7 +13,7 @@ from __future__ import annotations
 import collections
 import io
 import pickle
-from typing import Iterable, Optional, Tuple
+from typing import Optional
 
 import torch
 import torch.distributed as dist

@@ -29,7 +29,7 @@ from __future__ import annotations
 import struct
 import threading
 from concurrent.futures import Future, ThreadPoolExecutor
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import torch
 import torch.distributed as dist
@@ -38,7 +38,7 @@ from .. import common as C
 from ..common import telemetry
 from ..common.logging_util import get_logger
 from ..common.naming import partition_key
-from ..compression import BaseCompressor, Compressed
+from ..compression import BaseCompressor
 
 log = get_logger()
 
