@@ -9,9 +9,9 @@ Roles via ``BPS_ROLE`` (alias ``DMLC_ROLE``):
             launcher/launch.py:49-199 numactl pinning; no numactl
             dependency here — uses sched_setaffinity)
   server    run the native PS server (python -m byteps_amd.server)
-  scheduler no-op placeholder: rendezvous is static (BPS_SERVER_URIS +
-            MASTER_ADDR) — the reference needed a live scheduler because
-            ps-lite assigned node ids dynamically
+  scheduler run the rendezvous scheduler (byteps_amd.launcher.scheduler)
+            at BPS_ROOT_PORT; workers fall back to it when
+            BPS_SERVER_URIS is not set
 
 Usage:  bpslaunch python3 train.py [args...]
 """
@@ -121,9 +121,14 @@ def main() -> int:
         run_server()
         return 0
     if role == "scheduler":
-        print("byteps_amd: static rendezvous — scheduler role is a no-op; "
-              "set BPS_SERVER_URIS on workers", file=sys.stderr)
-        signal.pause()
+        from byteps_amd.launcher.scheduler import Scheduler
+        port = int(os.environ.get(
+            "BPS_ROOT_PORT", os.environ.get("DMLC_PS_ROOT_PORT", "9000")))
+        sched = Scheduler(port=port).start()
+        try:
+            signal.pause()
+        finally:
+            sched.stop()
         return 0
     print("unknown BPS_ROLE %r" % role, file=sys.stderr)
     return 2
