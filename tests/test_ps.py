@@ -336,3 +336,37 @@ def test_ps_per_param_compression_overrides(server):
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-4, atol=1e-5)
+
+
+def _sched_worker(rank, world):
+    import byteps_amd.torch as bps
+    bps.init()
+    t = torch.ones(64) * (rank + 1)
+    out = bps.push_pull(t, average=False, name="sched.pp")
+    ok = torch.allclose(out, torch.full((64,), float(sum(range(1, world + 1)))))
+    bps.shutdown()
+    return bool(ok)
+
+
+def test_ps_scheduler_discovery():
+    """Workers with no BPS_SERVER_URIS discover the server through the
+    rendezvous scheduler (reference DMLC_PS_ROOT_URI flow)."""
+    from byteps_amd.launcher.scheduler import Scheduler, register_server
+    from byteps_amd.ops import _core
+    sched = Scheduler(port=0, num_servers=1).start()
+    srv = _core.Server(0, 2, False)
+    srv.start()
+    try:
+        register_server("127.0.0.1", srv.port, "127.0.0.1", sched.port)
+        results = run_in_processes(
+            _sched_worker, 2,
+            extra_env={"BPS_FORCE_DISTRIBUTED": "1",
+                       "BPS_NUM_SERVER": "1",
+                       "BPS_ROOT_URI": "127.0.0.1",
+                       "BPS_ROOT_PORT": str(sched.port),
+                       "BPS_LOCAL_SIZE": "1",
+                       "LOCAL_WORLD_SIZE": "1"})
+        assert all(results)
+    finally:
+        srv.stop()
+        sched.stop()
