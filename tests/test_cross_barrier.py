@@ -61,3 +61,45 @@ def test_cross_barrier_world2():
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+def _cb_adam(rank, world, steps):
+    import byteps_amd.torch as bps
+    from byteps_amd.torch.cross_barrier import CrossBarrier
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.Adam(m.parameters(), lr=0.01)
+    cb = CrossBarrier(m, opt)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    for _ in range(steps):
+        cb.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        cb.step()
+    cb.synchronize()
+    out = [p.detach().clone() for p in m.parameters()]
+    cb.stop()
+    bps.shutdown()
+    return out
+
+
+def test_cross_barrier_adam_world2():
+    """Any-optimizer support: per-bucket Adam instances must match the
+    synchronous big-batch Adam run."""
+    world, steps = 2, 5
+    m = _make_model()
+    opt = torch.optim.Adam(m.parameters(), lr=0.01)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = torch.cat(xs), torch.cat(ys)
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        opt.step()
+    expected = [p.detach().clone() for p in m.parameters()]
+    results = run_in_processes(_cb_adam, world, steps)
+    for got in results:
+        for p_got, p_exp in zip(got, expected):
+            assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
