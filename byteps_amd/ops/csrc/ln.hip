@@ -15,7 +15,7 @@
 #include <hip/hip_bf16.h>
 
 #define BLOCK 256
-#define LN_RED_BLOCKS 256
+#define LN_RED_BLOCKS 1024
 
 namespace {
 
