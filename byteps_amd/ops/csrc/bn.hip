@@ -116,15 +116,26 @@ __global__ void bn_reduce_kernel(const bf16* __restrict__ x, long long M,
   }
 }
 
-// one-wave row reduction of a [RED_BLOCKS]-wide partial row
-__device__ inline float wave_row_sum(const float* __restrict__ row) {
+// one-wave reduction of the first `nblocks` entries of a partial row
+// (row stride is always RED_BLOCKS; small shapes fill fewer columns)
+__device__ inline float wave_row_sum(const float* __restrict__ row,
+                                     int nblocks) {
   const int lane = threadIdx.x & 63;
   float acc = 0.0f;
-#pragma unroll
-  for (int b = lane; b < RED_BLOCKS; b += 64) acc += row[b];
+  for (int b = lane; b < nblocks; b += 64) acc += row[b];
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
   return acc;  // valid in lane 0
+}
+
+// adaptive reduce grid: enough row-blocks to saturate HBM, no more than
+// the partial stride; small shapes stay shallow so the fold stays cheap
+inline int red_grid(long long M, int C) {
+  int groups = BLOCK / (C >> 3);
+  if (groups < 1) groups = 1;
+  long long need = (M + groups - 1) / groups;
+  if (need < 1) need = 1;
+  return (int)(need < RED_BLOCKS ? need : RED_BLOCKS);
 }
 
 // ---------------------------------------------------------------------------
@@ -139,11 +150,12 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partial,
                                    float* __restrict__ invstd_out,
                                    float* __restrict__ running_mean,
                                    float* __restrict__ running_var,
-                                   int update_running) {
+                                   int update_running, int nblocks) {
   int c = blockIdx.x;
   if (c >= C) return;
-  float sum = wave_row_sum(partial + (long long)c * RED_BLOCKS);
-  float sumsq = wave_row_sum(partial + (long long)(C + c) * RED_BLOCKS);
+  float sum = wave_row_sum(partial + (long long)c * RED_BLOCKS, nblocks);
+  float sumsq = wave_row_sum(partial + (long long)(C + c) * RED_BLOCKS,
+                             nblocks);
   if (threadIdx.x != 0) return;
   float n = (float)M;
   float mean = sum / n;
@@ -274,10 +286,10 @@ __global__ void bn_bwd_reduce_kernel(const bf16* __restrict__ x,
 
 // fold the transposed partials into sums2[2C] — one wave per row
 __global__ void bn_fold_kernel(const float* __restrict__ partial, int C,
-                               float* __restrict__ sums2) {
+                               float* __restrict__ sums2, int nblocks) {
   int row = blockIdx.x;
   if (row >= 2 * C) return;
-  float acc = wave_row_sum(partial + (long long)row * RED_BLOCKS);
+  float acc = wave_row_sum(partial + (long long)row * RED_BLOCKS, nblocks);
   if (threadIdx.x == 0) sums2[row] = acc;
 }
 
@@ -344,7 +356,7 @@ int bps_bn_red_blocks(void) { return RED_BLOCKS; }
 int bps_bn_reduce(const void* x, long long M, int C, void* partial,
                   void* stream) {
   if ((C & 7) || C > 2048) return -1;
-  hipLaunchKernelGGL(bn_reduce_kernel, dim3(RED_BLOCKS), dim3(BLOCK), 0,
+  hipLaunchKernelGGL(bn_reduce_kernel, dim3(red_grid(M, C)), dim3(BLOCK), 0,
                      STREAM, (const bf16*)x, M, C, (float*)partial);
   return (int)hipGetLastError();
 }
@@ -357,7 +369,7 @@ int bps_bn_finalize(const void* sums, long long M, int C, float eps,
                      STREAM, (const float*)sums, M, C, eps, momentum,
                      (float*)mean_out, (float*)invstd_out,
                      (float*)running_mean, (float*)running_var,
-                     update_running);
+                     update_running, red_grid(M, C));
   return (int)hipGetLastError();
 }
 
@@ -388,14 +400,15 @@ int bps_bn_bwd_reduce(const void* x, const void* dy, const void* mask,
                       void* stream) {
   if ((C & 7) || C > 2048) return -1;
   if (relu && !mask) return -2;
+  int g = red_grid(M, C);
   if (relu)
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(RED_BLOCKS),
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<true>), dim3(g),
                        dim3(BLOCK), 0, STREAM, (const bf16*)x,
                        (const bf16*)dy, (const unsigned char*)mask, M, C,
                        (const float*)mean, (const float*)invstd,
                        (float*)partial);
   else
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(RED_BLOCKS),
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<false>), dim3(g),
                        dim3(BLOCK), 0, STREAM, (const bf16*)x,
                        (const bf16*)dy, (const unsigned char*)mask, M, C,
                        (const float*)mean, (const float*)invstd,
@@ -403,9 +416,11 @@ int bps_bn_bwd_reduce(const void* x, const void* dy, const void* mask,
   return (int)hipGetLastError();
 }
 
-int bps_bn_fold(const void* partial, int C, void* sums2, void* stream) {
+int bps_bn_fold(const void* partial, long long M, int C, void* sums2,
+                void* stream) {
   hipLaunchKernelGGL(bn_fold_kernel, dim3(2 * C), dim3(64), 0, STREAM,
-                     (const float*)partial, C, (float*)sums2);
+                     (const float*)partial, C, (float*)sums2,
+                     red_grid(M, C));
   return (int)hipGetLastError();
 }
 

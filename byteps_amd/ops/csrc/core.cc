@@ -63,7 +63,8 @@ int bps_bn_bwd_reduce(const void* x, const void* dy, const void* mask,
                       long long M, int C, const void* mean,
                       const void* invstd, void* partial, int relu,
                       void* stream);
-int bps_bn_fold(const void* partial, int C, void* sums2, void* stream);
+int bps_bn_fold(const void* partial, long long M, int C, void* sums2,
+                void* stream);
 int bps_bn_red_blocks(void);
 
 // -- fused layernorm (ln.hip) ----------------------------------------------
@@ -75,7 +76,8 @@ int bps_ln_fwd(const void* x, const void* gamma, const void* beta, void* y,
 int bps_ln_bwd(const void* x, const void* dy, const void* gamma,
                const void* mean, const void* invstd, void* dx, long long M,
                int C, void* partial, void* stream);
-int bps_ln_fold(const void* partial, int C, void* sums2, void* stream);
+int bps_ln_fold(const void* partial, long long M, int C, void* sums2,
+                void* stream);
 int bps_bn_bwd_apply(const void* x, const void* dy, const void* mask,
                      void* dx,
                      void* dres, long long M, int C, const void* mean,
@@ -231,9 +233,9 @@ PYBIND11_MODULE(_core, m) {
                         uintptr_t s) {
     check(bps_bn_reduce(CP(x), M, C, P(sums), P(s)), "bps_bn_reduce");
   });
-  m.def("bn_fold", [](uintptr_t partial, int C, uintptr_t sums2,
+  m.def("bn_fold", [](uintptr_t partial, int64_t M, int C, uintptr_t sums2,
                       uintptr_t s) {
-    check(bps_bn_fold(CP(partial), C, P(sums2), P(s)), "bps_bn_fold");
+    check(bps_bn_fold(CP(partial), M, C, P(sums2), P(s)), "bps_bn_fold");
   });
   m.def("bn_finalize",
         [](uintptr_t sums, int64_t M, int C, float eps, float momentum,
@@ -289,9 +291,9 @@ PYBIND11_MODULE(_core, m) {
                            P(dx), M, C, P(partial), P(s)),
                 "bps_ln_bwd");
         });
-  m.def("ln_fold", [](uintptr_t partial, int C, uintptr_t sums2,
+  m.def("ln_fold", [](uintptr_t partial, int64_t M, int C, uintptr_t sums2,
                       uintptr_t s) {
-    check(bps_ln_fold(CP(partial), C, P(sums2), P(s)), "bps_ln_fold");
+    check(bps_ln_fold(CP(partial), M, C, P(sums2), P(s)), "bps_ln_fold");
   });
 
   // CPU reducer / codecs

@@ -209,21 +209,31 @@ __global__ void ln_bwd_kernel(const bf16* __restrict__ x,
   }
 }
 
-__device__ inline float wave_row_sum_ln(const float* __restrict__ row) {
+__device__ inline float wave_row_sum_ln(const float* __restrict__ row,
+                                        int nblocks) {
   const int lane = threadIdx.x & 63;
   float acc = 0.0f;
-#pragma unroll
-  for (int b = lane; b < LN_RED_BLOCKS; b += 64) acc += row[b];
+  for (int b = lane; b < nblocks; b += 64) acc += row[b];
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
   return acc;
 }
 
+// adaptive grid matching bn.hip's red_grid (stride stays LN_RED_BLOCKS)
+inline int ln_red_grid(long long M, int C) {
+  int groups = BLOCK / (C >> 3);
+  if (groups < 1) groups = 1;
+  long long need = (M + groups - 1) / groups;
+  if (need < 1) need = 1;
+  return (int)(need < LN_RED_BLOCKS ? need : LN_RED_BLOCKS);
+}
+
 __global__ void ln_fold_kernel(const float* __restrict__ partial, int C,
-                               float* __restrict__ sums2) {
+                               float* __restrict__ sums2, int nblocks) {
   int row = blockIdx.x;
   if (row >= 2 * C) return;
-  float acc = wave_row_sum_ln(partial + (long long)row * LN_RED_BLOCKS);
+  float acc = wave_row_sum_ln(partial + (long long)row * LN_RED_BLOCKS,
+                              nblocks);
   if (threadIdx.x == 0) sums2[row] = acc;
 }
 
@@ -262,17 +272,19 @@ int bps_ln_bwd(const void* x, const void* dy, const void* gamma,
                const void* mean, const void* invstd, void* dx, long long M,
                int C, void* partial, void* stream) {
   if (!ln_supported(C)) return -1;
-  hipLaunchKernelGGL(ln_bwd_kernel, dim3(LN_RED_BLOCKS), dim3(BLOCK), 0,
-                     STREAM, (const bf16*)x, (const bf16*)dy,
+  hipLaunchKernelGGL(ln_bwd_kernel, dim3(ln_red_grid(M, C)), dim3(BLOCK),
+                     0, STREAM, (const bf16*)x, (const bf16*)dy,
                      (const float*)gamma, (const float*)mean,
                      (const float*)invstd, (bf16*)dx, M, C,
                      (float*)partial);
   return (int)hipGetLastError();
 }
 
-int bps_ln_fold(const void* partial, int C, void* sums2, void* stream) {
+int bps_ln_fold(const void* partial, long long M, int C, void* sums2,
+                void* stream) {
   hipLaunchKernelGGL(ln_fold_kernel, dim3(2 * C), dim3(64), 0, STREAM,
-                     (const float*)partial, C, (float*)sums2);
+                     (const float*)partial, C, (float*)sums2,
+                     ln_red_grid(M, C));
   return (int)hipGetLastError();
 }
 

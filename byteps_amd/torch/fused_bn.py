@@ -90,7 +90,7 @@ class _FusedBNFunction(torch.autograd.Function):
                            mean.data_ptr(), invstd.data_ptr(),
                            partial.data_ptr(), int(ctx.relu), s)
         sums2 = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-        core.bn_fold(partial.data_ptr(), C, sums2.data_ptr(), s)
+        core.bn_fold(partial.data_ptr(), M, C, sums2.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         core.bn_bwd_apply(x.data_ptr(), dy.data_ptr(), mask_ptr,

@@ -53,7 +53,7 @@ class _FusedLNFunction(torch.autograd.Function):
                     mean.data_ptr(), invstd.data_ptr(), dx.data_ptr(), M, C,
                     partial.data_ptr(), _stream(x))
         sums2 = torch.empty(2 * C, dtype=torch.float32, device=x.device)
-        core.ln_fold(partial.data_ptr(), C, sums2.data_ptr(), _stream(x))
+        core.ln_fold(partial.data_ptr(), M, C, sums2.data_ptr(), _stream(x))
         dbeta = sums2[:C]
         dgamma = sums2[C:]
         return dx, dgamma, dbeta, None

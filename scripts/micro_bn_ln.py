@@ -72,7 +72,7 @@ def bench_bn(N, C, H, W):
                                      invstd.data_ptr(), partial.data_ptr(),
                                      1, st),
           nbytes * 2)
-    core.bn_fold(partial.data_ptr(), C, sums2.data_ptr(), st)
+    core.bn_fold(partial.data_ptr(), M, C, sums2.data_ptr(), st)
     timed(tag + " bwd_apply(+dres,relu)",
           lambda: core.bn_bwd_apply(x.data_ptr(), dy.data_ptr(),
                                     mask.data_ptr(), dx.data_ptr(),
