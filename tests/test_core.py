@@ -104,3 +104,24 @@ def test_mixed_mode_weighted_assignment(monkeypatch):
     # colocated server 0 gets ~half the standalone server's bytes
     ratio = a.load[0] / a.load[1]
     assert 0.4 < ratio < 0.65, ratio
+
+
+def test_ops_cpu_wrappers():
+    import byteps_amd.ops as K
+    t = torch.arange(16, dtype=torch.float32)
+    K.scale_(t, 0.5)
+    assert torch.allclose(t, torch.arange(16, dtype=torch.float32) * 0.5)
+    y = torch.ones(16)
+    K.axpy_(y, t, 2.0)
+    assert torch.allclose(y, 1 + 2 * t)
+    assert abs(K.norm(t, "l1").item() - t.abs().sum().item()) < 1e-4
+    assert abs(K.norm(t, "l2").item() - t.norm().item()) < 1e-4
+    assert abs(K.norm(t, "max").item() - t.abs().max().item()) < 1e-6
+
+
+def test_telemetry_window_decay():
+    from byteps_amd.common.telemetry import SpeedMeter
+    m = SpeedMeter()
+    m.record(1_000_000)
+    _, mbps = m.speed()
+    assert mbps >= 0.0
