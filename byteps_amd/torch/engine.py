@@ -459,6 +459,21 @@ def _suspend_engines() -> None:
 def _resume_engines() -> None:
     for e in _engines:
         e._attach_hooks()
+        # topology may have changed — refresh the averaging divisor.
+        # (Bucket layout is NOT rebuilt: if the new world no longer
+        # divides the bucket alignment the PS shard path would truncate,
+        # so warn loudly — full elastic re-bucketing is future work; the
+        # reference likewise only re-declared keys, operations.cc:96-119.)
+        new_world = dist.get_world_size() if dist.is_initialized() else 1
+        if new_world != e.world:
+            e.world = new_world
+            for b in e.buckets:
+                if b.buffer.numel() % max(1, new_world) != 0:
+                    log.warning(
+                        "resume: bucket %d (%d elems) not divisible by new "
+                        "world %d — PS sharding disabled would be unsafe; "
+                        "recreate the optimizer/DDP for clean re-bucketing",
+                        b.plan.index, b.buffer.numel(), new_world)
         e._ps = None
         if C._state.ps_enabled:
             from . import ps_pipeline
