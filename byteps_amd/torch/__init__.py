@@ -210,8 +210,6 @@ def broadcast_optimizer_state(optimizer, root_rank: int = 0) -> None:
     torch/__init__.py:302-424): tensor state entries are broadcast
     in-place; scalar entries (e.g. ``step``) travel pickled."""
     _C._require_init()
-    if isinstance(optimizer, _DistributedOptimizer.__bases__[0].__class__):
-        pass
     if _C.size() <= 1 or not dist.is_initialized():
         return
 
