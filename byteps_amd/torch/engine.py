@@ -74,6 +74,12 @@ class Bucket:
     # per-step state
     ready_count: int = 0
     issued: bool = False
+    # engine-step number of the last issue.  Unlike ``issued`` (cleared by
+    # reset(), which the cross-barrier poller calls as soon as a bucket's
+    # update lands), the stamp survives until the next step — flush()
+    # keys on it so a bucket can never be double-issued within one step
+    # when the poller wins the race against step()'s flush.
+    stamp: int = -1
     work: Optional[object] = None             # dist Work handle
     ps_ticket: Optional[object] = None        # PS pipeline ticket
 
@@ -316,9 +322,10 @@ class GradEngine:
             self._issue(b)
 
     def _issue(self, b: Bucket) -> None:
-        if b.issued:
+        if b.issued or b.stamp == self._step:
             return
         b.issued = True
+        b.stamp = self._step
         self._inflight_bytes += b.nbytes
         if C._state.tracer is not None:
             C._state.tracer.begin(b.declared_key, "comm", self._step)
@@ -378,7 +385,7 @@ class GradEngine:
         (their spans hold zeros — contributing zero to the sum is correct)."""
         with self._lock:
             for b in self.buckets:
-                if not b.issued:
+                if not b.issued and b.stamp != self._step:
                     heapq.heappush(self._pending, (-b.priority, b.plan.index))
             saved_credit = self._credit
             self._credit = 0          # credits never hold back a flush
