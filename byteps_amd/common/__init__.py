@@ -41,6 +41,8 @@ class _State:
         self.owns_process_group = False
         self.kv = None           # KV client handle (PS mode)
         self.tracer = None
+        self.base_master_port: Optional[int] = None
+        self.resume_count = 0
 
     @property
     def ps_enabled(self) -> bool:
@@ -91,6 +93,8 @@ def init(lazy: bool = True, backend: Optional[str] = None) -> None:
     if size > 1 and not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")
+        if _state.base_master_port is None:
+            _state.base_master_port = int(os.environ["MASTER_PORT"])
         dist.init_process_group(
             backend=backend or _default_backend(),
             rank=rank,
@@ -164,6 +168,13 @@ def resume(num_workers: int, num_servers: int,
     os.environ["BPS_NUM_SERVER"] = str(num_servers)
     if global_rank is not None:
         os.environ["RANK"] = str(global_rank)
+    # rotate the rendezvous port deterministically: re-binding the old
+    # MASTER_PORT races the previous TCPStore's teardown (observed hangs
+    # under load); every rank derives the same next port
+    _state.resume_count += 1
+    if _state.base_master_port is not None:
+        os.environ["MASTER_PORT"] = str(
+            _state.base_master_port + _state.resume_count)
     _state.registry.redeclare_all()
     init()
     from ..torch import engine as torch_engine
