@@ -350,18 +350,35 @@ class Server {
     MsgHeader h = req;
     h.len = 0;
     h.aux = ~0ULL;  // error marker
-    h.op = req.op == kPull ? kPullReply : kPushReply;
+    h.op = req.op == kPull ? kPullReply
+         : req.op == kInit ? kInitReply : kPushReply;
     conn->send(h, nullptr);
   }
 
   void handle_init(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
                    const std::vector<char>& payload) {
+    bool mismatch = false;
     if (payload.size() >= 16) {
       InitPayload ip{};
       std::memcpy(&ip, payload.data(),
                   std::min(payload.size(), sizeof(ip)));
       std::lock_guard<std::mutex> lk(keys_mu_);
       auto& slot = keys_[h.key];
+      if (slot) {
+        // re-init (elastic resume): the config must match the stored
+        // state — stale sizes would later cause out-of-bounds merges.
+        // A changed world (expected) or levels is adopted; a changed
+        // nelem/codec is an error the worker must see.
+        std::lock_guard<std::mutex> klk(slot->mu);
+        if (slot->nelem != ip.nelem || slot->codec != cmd_codec(h.cmd)) {
+          mismatch = true;
+        } else {
+          slot->expected = std::max(1u, ip.expected);
+          slot->levels = std::max(1u, ip.levels);
+          slot->async_mode = cmd_async(h.cmd);
+          slot->server_ef = (ip.flags & 1u) != 0;
+        }
+      }
       if (!slot) {
         slot = std::make_unique<KeyState>();
         slot->nelem = ip.nelem;
@@ -381,6 +398,10 @@ class Server {
         engine_of_[h.key] = best;
         engine_load_[best] += (int64_t)ip.nelem;
       }
+    }
+    if (mismatch) {
+      reply_err(conn, h);
+      return;
     }
     MsgHeader r = h;
     r.op = kInitReply;

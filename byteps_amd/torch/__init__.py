@@ -84,11 +84,15 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             self._async_worker = AsyncPSWorker(named)
             return
 
-        grad_dtype = (torch.bfloat16 if compression is Compression.bf16
+        # Compression.fp16/bf16 narrows only the WIRE (the all-reduce
+        # scratch), never the gradient accumulation dtype — reference
+        # torch/compression.py:34-76 compressed at enqueue and
+        # decompressed at synchronize, leaving autograd untouched.
+        comm_dtype = (torch.bfloat16 if compression is Compression.bf16
                       else torch.float16 if compression is Compression.fp16
                       else None)
         self._engine = register_engine(GradEngine(
-            named, process_group=process_group, grad_dtype=grad_dtype,
+            named, process_group=process_group, comm_dtype=comm_dtype,
             compression_params=compression_params))
         if backward_passes_per_step > 1:
             self._engine.set_sync_enabled(False)
@@ -115,8 +119,12 @@ class _DistributedOptimizer(torch.optim.Optimizer):
             return None
         self._pass_count = 0
         if self.backward_passes_per_step > 1:
-            # one full accumulation window done: sync the accumulated grads
+            # one full accumulation window done: sync the accumulated grads.
+            # Split params accumulated into their PRIVATE grad while sync
+            # was off — refresh the bucket views first, or the flush would
+            # all-reduce the stale spans.
             self._engine.set_sync_enabled(True)
+            self._engine.sync_split_grads()
             self._engine.flush()
         self._engine.synchronize()
         loss = super(self.__class__, self).step(closure)

@@ -59,16 +59,38 @@ class CrossBarrier:
             for p in group["params"]:
                 self._group_of[id(p)] = group
 
-        # one optimizer instance per bucket over exactly its params
-        opt_cls = type(optimizer)
-        self._bucket_opts: Dict[int, torch.optim.Optimizer] = {}
-        for b in self._engine.buckets:
+        # Params split across buckets (numel > partition size) need special
+        # handling: their p.grad is PRIVATE (not a bucket view), refreshed
+        # from the averaged shard views only once ALL containing buckets
+        # completed — and stepped exactly once, by a dedicated optimizer.
+        eng = self._engine
+        self._split_pidx_of = {id(eng.params[pidx]): pidx
+                               for pidx in eng._split_params}
+
+        def _mk_opt(opt_cls, params):
             groups = []
-            for p in dict.fromkeys(b.params):
+            for p in params:
                 g = self._group_of.get(id(p))
                 hyper = {k: v for k, v in (g or {}).items() if k != "params"}
                 groups.append({"params": [p], **hyper})
-            self._bucket_opts[b.plan.index] = opt_cls(groups)
+            return opt_cls(groups)
+
+        # one optimizer instance per bucket over exactly its NON-split
+        # params; one per split param (stepped by its last bucket)
+        opt_cls = type(optimizer)
+        self._bucket_opts: Dict[int, torch.optim.Optimizer] = {}
+        for b in self._engine.buckets:
+            whole = [p for p in dict.fromkeys(b.params)
+                     if id(p) not in self._split_pidx_of]
+            if whole:
+                self._bucket_opts[b.plan.index] = _mk_opt(opt_cls, whole)
+        self._split_opts: Dict[int, torch.optim.Optimizer] = {}
+        self._split_total: Dict[int, int] = {}
+        self._split_remaining: Dict[int, int] = {}
+        for pidx, bks in eng._split_params.items():
+            self._split_opts[pidx] = _mk_opt(opt_cls, [eng.params[pidx]])
+            self._split_total[pidx] = len(bks)
+            self._split_remaining[pidx] = len(bks)
 
         # per-parameter "updated" events; forward pre-hooks block on them
         self._events: Dict[int, threading.Event] = {}
@@ -107,7 +129,10 @@ class CrossBarrier:
             except Exception:
                 log.exception("cross-barrier poller failed")
             for p in bucket.params:
-                self._events[id(p)].set()
+                # split params unlock only when their LAST bucket lands
+                # (_finish_bucket sets those events itself)
+                if id(p) not in self._split_pidx_of:
+                    self._events[id(p)].set()
             with self._pending_lock:
                 self._pending -= 1
                 if self._pending == 0:
@@ -123,16 +148,41 @@ class CrossBarrier:
                 bucket.buffer.copy_(eng._wire_scratch[bucket.plan.index])
         if eng.average and eng.world > 1 and not eng.prescale:
             bucket.buffer.div_(eng.world)
+        opt = self._bucket_opts.get(bucket.plan.index)
+        if opt is not None:
+            self._refresh_hyper(opt)
+            opt.step()
+        # split params: when the last containing bucket lands, refresh
+        # p.grad from the averaged shard views and step exactly once
+        for p in bucket.params:
+            pidx = self._split_pidx_of.get(id(p))
+            if pidx is None:
+                continue
+            self._split_remaining[pidx] -= 1
+            if self._split_remaining[pidx] > 0:
+                continue
+            self._split_remaining[pidx] = self._split_total[pidx]
+            flat = p.grad.reshape(-1)
+            off = 0
+            for b in eng._split_params[pidx]:
+                for q, g in zip(b.params, b.grads):
+                    if q is p:
+                        flat.narrow(0, off, g.numel()).copy_(g)
+                        off += g.numel()
+            sopt = self._split_opts[pidx]
+            self._refresh_hyper(sopt)
+            sopt.step()
+            self._events[id(p)].set()
+        bucket.reset()
+
+    def _refresh_hyper(self, opt: torch.optim.Optimizer) -> None:
         # refresh live hyperparams (lr schedules) from the user optimizer
-        opt = self._bucket_opts[bucket.plan.index]
         for g in opt.param_groups:
             src = self._group_of.get(id(g["params"][0]))
             if src:
                 for k, v in src.items():
                     if k != "params":
                         g[k] = v
-        opt.step()
-        bucket.reset()
 
     # -- forward gating ------------------------------------------------------
 

@@ -116,6 +116,7 @@ class GradEngine:
         grad_dtype: Optional[torch.dtype] = None,
         prescale: bool = False,
         compression_params: Optional[dict] = None,
+        comm_dtype: Optional[torch.dtype] = None,
     ) -> None:
         self.compression_params = dict(compression_params or {})
         C._require_init()
@@ -242,14 +243,18 @@ class GradEngine:
                 self._split_params[pidx] = bks
 
         # optional reduced-precision wire for the RCCL collectives
-        # (BPS_COMM_DTYPE=bf16): cast bucket → persistent bf16 scratch,
+        # (``comm_dtype`` argument — set by Compression.fp16/bf16 — or
+        # BPS_COMM_DTYPE=bf16): cast bucket → persistent 16-bit scratch,
         # all-reduce the scratch, cast back at synchronize.  Halves xGMI
-        # bytes with bf16's fp32 exponent range — the role the
-        # reference's fp16 Compression played (torch/compression.py).
-        from ..common.config import env_str
-        wire = env_str("BPS_COMM_DTYPE", default="").lower()
-        self.comm_dtype = torch.bfloat16 if wire in ("bf16", "bfloat16") \
-            else torch.float16 if wire in ("fp16", "half") else None
+        # bytes; gradients still ACCUMULATE at full precision — only the
+        # wire is narrowed, matching the reference's Compression semantics
+        # (torch/compression.py:34-76).
+        if comm_dtype is None:
+            from ..common.config import env_str
+            wire = env_str("BPS_COMM_DTYPE", default="").lower()
+            comm_dtype = torch.bfloat16 if wire in ("bf16", "bfloat16") \
+                else torch.float16 if wire in ("fp16", "half") else None
+        self.comm_dtype = comm_dtype
         if self.comm_dtype is not None and dt == self.comm_dtype:
             self.comm_dtype = None
         self._wire_scratch: Dict[int, torch.Tensor] = {}
@@ -380,6 +385,21 @@ class GradEngine:
         parallel/distributed.py:184-207)."""
         self._sync_enabled = enabled
 
+    def sync_split_grads(self) -> None:
+        """Copy every split param's private (accumulated) grad into its
+        bucket views.  Needed before flush() whenever the hooks ran with
+        sync disabled (gradient accumulation): the hooks early-return in
+        that state, so the views still hold the previous round's bytes."""
+        for pidx, bks in self._split_params.items():
+            p = self.params[pidx]
+            flat = p.grad.reshape(-1)
+            off = 0
+            for b in bks:
+                for q, g in zip(b.params, b.grads):
+                    if q is p:
+                        g.copy_(flat.narrow(0, off, g.numel()))
+                        off += g.numel()
+
     def flush(self) -> None:
         """Force-issue buckets whose params produced no grad this step
         (their spans hold zeros — contributing zero to the sum is correct)."""
@@ -409,8 +429,10 @@ class GradEngine:
         if self.average and self.world > 1 and not self.prescale:
             torch._foreach_div_([b.buffer for b in self.buckets],
                                 float(self.world))
-        if self.average and self._split_params:
-            # split params read back their averaged grads
+        if self._split_params:
+            # split params read back their reduced (averaged) grads —
+            # unconditionally: with average=False the SUM must still
+            # reach p.grad
             for pidx, bks in self._split_params.items():
                 p = self.params[pidx]
                 flat = p.grad.reshape(-1)
@@ -450,6 +472,8 @@ def _shutdown_engine() -> None:
     for e in _engines:
         try:
             e.detach()
+            if e._ps is not None:
+                e._ps.close()
         except Exception:
             pass
     _engines.clear()
@@ -461,6 +485,12 @@ def _suspend_engines() -> None:
     common/operations.cc:96-107)."""
     for e in _engines:
         e.detach()
+        if e._ps is not None:
+            try:
+                e._ps.close()
+            except Exception:
+                pass
+            e._ps = None
 
 
 def _resume_engines() -> None:

@@ -19,9 +19,15 @@ shm even on one node, common/core_loops.cc:378-443).
 Collective-ordering rule: RCCL collectives on one communicator must be
 issued in the same order on every rank.  Reduce-scatters are issued from
 autograd hooks (same backward graph ⇒ same order — the same contract
-torch DDP relies on); all-gathers are issued in fixed bucket-index order
-during synchronize().  The middle (KV) section runs in a thread pool with
-per-task side streams and completes in any order.
+torch DDP relies on) on the node group; trailing all-gathers are issued
+by ONE dedicated issuer thread, in submit (hook) order, on a SEPARATE
+communicator (``ag_group``) — so the two phases can never interleave
+differently across ranks, whichever thread (synchronize() or the
+cross-barrier poller) consumes the results.  The middle (KV) section
+runs in a thread pool with per-task side streams and completes in any
+order; push and pull are in flight together (full duplex — reference
+docs/faq.md:23-25), the server defers the pull reply until the round's
+merge is complete.
 """
 
 from __future__ import annotations
@@ -80,14 +86,19 @@ class _Staging:
 
 
 class Ticket:
-    __slots__ = ("future", "bucket", "shard", "done_event", "reply_view")
+    __slots__ = ("future", "bucket", "shard", "done_event", "reply_view",
+                 "seq", "ag_done", "ag_work", "error")
 
-    def __init__(self, future: Future, bucket, shard):
+    def __init__(self, future: Optional[Future], bucket, shard):
         self.future = future
         self.bucket = bucket
         self.shard = shard
-        self.done_event = None
+        self.done_event = None          # HIP event: KV H2D complete
         self.reply_view = None
+        self.seq = -1                   # submit order (same on all ranks)
+        self.ag_done = threading.Event()  # issuer finished this ticket
+        self.ag_work = None             # dist Work of the trailing collective
+        self.error = None               # exception raised on the issuer
 
 
 def _kv_client():
@@ -141,20 +152,41 @@ class PSPipeline:
         self._lock = threading.Lock()
         self._staging: Dict[int, _Staging] = {}
 
-        # intra-node subgroup (reduce-scatter / all-gather scope).
+        # intra-node subgroups: one for the leading reduce-scatters
+        # (node_group, autograd-hook thread) and a SEPARATE communicator
+        # for the trailing all-gathers (ag_group, issuer thread) — two
+        # threads must never issue onto one communicator (reference kept
+        # per-purpose NCCL rings too, nccl_manager.cc:83-84)
         self.node_group = None
+        self.ag_group = None
         if self.world > 1 and dist.is_initialized():
             if self.num_nodes > 1:
                 for nid in range(self.num_nodes):
                     ranks = list(range(nid * self.local_size,
                                        (nid + 1) * self.local_size))
                     g = dist.new_group(ranks)
+                    g2 = dist.new_group(ranks)
                     if nid == self.node_id:
                         self.node_group = g
+                        self.ag_group = g2
             else:
                 self.node_group = None  # default group == node group
+                self.ag_group = dist.new_group(list(range(self.world)))
         self.node_world = dist.get_world_size(self.node_group) \
             if (self.world > 1 and dist.is_initialized()) else 1
+
+        # ordered issuer: tickets enter in submit (hook) order; the issuer
+        # waits for each ticket's KV section, then issues its trailing
+        # collective — strict FIFO, so issue order is identical on every
+        # rank by construction
+        import queue as _q
+        self._issue_q: "_q.Queue[Optional[Ticket]]" = _q.Queue()
+        self._seq = 0
+        self._ag_stream = None
+        self._closed = False
+        self._issuer = threading.Thread(target=self._issue_loop, daemon=True,
+                                        name="bps-ps-issuer")
+        self._issuer.start()
 
         # engine-wide defaults; each bucket carries its own resolved
         # config (per-parameter overrides, reference byteps_* attrs)
@@ -225,29 +257,32 @@ class PSPipeline:
         cmd = _make_cmd(codec, 0, self.cfg.enable_async)
         t = self.kv.submit(ki.server, _OP_INIT, ki.key, buf.data_ptr(),
                            len(payload), 0, 0, cmd, 0)
-        self.kv.wait(t)
+        _len, aux = self.kv.wait(t)
+        if aux == 0xFFFFFFFFFFFFFFFF:
+            raise RuntimeError(
+                "PS server rejected re-init of key %d: nelem/codec differ "
+                "from the stored state (elastic resume with a changed "
+                "config needs new keys — recreate the optimizer)" % ki.key)
         ki.initialized = True
 
     # -- main entry ---------------------------------------------------------
 
     def submit(self, bucket) -> Ticket:
         """Called from the autograd hook (deterministic order).  Issues the
-        intra-node reduce-scatter inline, then hands the KV section to the
-        pool."""
+        intra-node reduce-scatter inline, hands the KV section to the
+        pool, and registers the ticket with the ordered issuer."""
         ki = self._key_info(bucket)
         buf = bucket.buffer
         dev = buf.device
 
+        skip_kv = False
         if self.node_world > 1 and self.reduce_roots:
             root_local = bucket.declared_key % self.node_world
             root_global = self.node_id * self.node_world + root_local
             dist.reduce(buf, dst=root_global, group=self.node_group)
-            if self.local_rank != root_local:
-                # non-root: nothing to push; wait() broadcasts the result
-                fut = Future()
-                fut.set_result((None, None))
-                return Ticket(fut, bucket, buf)
             shard = buf
+            # non-root: nothing to push; the issuer broadcasts the result
+            skip_kv = self.local_rank != root_local
         elif self.node_world > 1:
             per = buf.numel() // self.node_world
             shard = buf.narrow(0, self.local_rank * per, per)
@@ -263,8 +298,18 @@ class PSPipeline:
             rs_event = torch.cuda.Event()
             rs_event.record(torch.cuda.current_stream(dev))
 
-        fut = self.pool.submit(self._kv_section, bucket, ki, shard, rs_event)
-        return Ticket(fut, bucket, shard)
+        if skip_kv:
+            fut = Future()
+            fut.set_result((None, None))
+            ticket = Ticket(fut, bucket, shard)
+        else:
+            fut = self.pool.submit(self._kv_section, bucket, ki, shard,
+                                   rs_event)
+            ticket = Ticket(fut, bucket, shard)
+        ticket.seq = self._seq
+        self._seq += 1
+        self._issue_q.put(ticket)
+        return ticket
 
     def _kv_section(self, bucket, ki: _KeyInfo, shard: torch.Tensor,
                     rs_event) -> tuple:
@@ -308,17 +353,22 @@ class PSPipeline:
         _tr("compress+d2h", False)
         codec = comp.codec if comp is not None else 0
         cmd = _make_cmd(codec, 0, self.cfg.enable_async)
+        # full duplex: push and pull are both in flight — the server
+        # defers the pull reply until this round's merge is complete
+        # (version gate), so the pull can be posted immediately and its
+        # latency overlaps other buckets' pushes (reference
+        # docs/faq.md:23-25)
         _tr("push", True)
         t_push = self.kv.submit(ki.server, _OP_PUSH, ki.key,
                                 st.send.data_ptr(), nbytes, 0, 0, cmd,
                                 push_aux)
+        t_pull = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
+                                st.recv.data_ptr(), st.recv.numel(), cmd,
+                                ki.round)
         self.kv.wait(t_push)
         _tr("push", False)
         telemetry.record(nbytes)
         _tr("pull", True)
-        t_pull = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
-                                st.recv.data_ptr(), st.recv.numel(), cmd,
-                                ki.round)
         reply_len, _ver = self.kv.wait(t_pull)
         _tr("pull", False)
         if reply_len > st.recv.numel():
@@ -356,29 +406,76 @@ class PSPipeline:
         _tr("h2d+decompress", False)
         return done_event, None
 
-    def wait(self, ticket: Ticket) -> None:
-        """Called from synchronize() in fixed bucket order: join the KV
-        future, chain the side stream into the main stream, and issue the
-        trailing all-gather (deterministic order)."""
+    def _issue_loop(self) -> None:
+        """Single issuer thread: strict FIFO over tickets (= submit order,
+        identical on every rank), so the trailing collectives on ag_group
+        can never interleave differently across ranks — whichever thread
+        later consumes the results."""
+        while True:
+            ticket = self._issue_q.get()
+            if ticket is None:
+                return
+            try:
+                self._issue_one(ticket)
+            except Exception as e:   # surface on the waiter, don't die
+                ticket.error = e
+            ticket.ag_done.set()
+
+    def _issue_one(self, ticket: Ticket) -> None:
         done_event, _ = ticket.future.result()
+        ticket.done_event = done_event
         buf = ticket.bucket.buffer
-        if done_event is not None:
-            torch.cuda.current_stream(buf.device).wait_event(done_event)
-        if self.reduce_roots and self.node_world > 1:
-            root_local = ticket.bucket.declared_key % self.node_world
-            root_global = self.node_id * self.node_world + root_local
-            dist.broadcast(buf, src=root_global, group=self.node_group)
-            return
-        if self.node_world > 1:
-            if dist.get_backend(self.node_group) == "nccl":
-                dist.all_gather_into_tensor(buf, ticket.shard,
-                                            group=self.node_group)
+        on_gpu = buf.is_cuda
+        if self.node_world <= 1:
+            return                      # nothing collective to issue
+        if on_gpu and self._ag_stream is None:
+            self._ag_stream = torch.cuda.Stream(buf.device)
+        stream_ctx = torch.cuda.stream(self._ag_stream) if on_gpu \
+            else _null_ctx()
+        with stream_ctx:
+            if done_event is not None:
+                self._ag_stream.wait_event(done_event)
+                ticket.done_event = None    # consumed by the ag chain
+            if self.reduce_roots:
+                root_local = ticket.bucket.declared_key % self.node_world
+                root_global = self.node_id * self.node_world + root_local
+                ticket.ag_work = dist.broadcast(
+                    buf, src=root_global, group=self.ag_group, async_op=True)
+            elif dist.get_backend(self.ag_group) == "nccl":
+                ticket.ag_work = dist.all_gather_into_tensor(
+                    buf, ticket.shard, group=self.ag_group, async_op=True)
             else:
                 # chunks are contiguous views of buf — all_gather fills
                 # the bucket in place
                 chunks = list(buf.chunk(self.node_world))
-                dist.all_gather(chunks, ticket.shard.contiguous(),
-                                group=self.node_group)
+                ticket.ag_work = dist.all_gather(
+                    chunks, ticket.shard.contiguous(), group=self.ag_group,
+                    async_op=True)
+
+    def wait(self, ticket: Ticket) -> None:
+        """Called from synchronize() or the cross-barrier poller (any
+        order): join the issuer, then chain the result into the caller's
+        current stream."""
+        ticket.ag_done.wait()
+        if ticket.error is not None:
+            raise ticket.error
+        if ticket.future is not None:
+            ticket.future.result()      # re-raise KV-section exceptions
+        if ticket.ag_work is not None:
+            # Work.wait() blocks (gloo) / makes the current stream wait
+            # for the collective (nccl)
+            ticket.ag_work.wait()
+        elif ticket.done_event is not None:
+            torch.cuda.current_stream(
+                ticket.bucket.buffer.device).wait_event(ticket.done_event)
+
+    def close(self) -> None:
+        if self._closed:
+            return
+        self._closed = True
+        self._issue_q.put(None)
+        self._issuer.join(timeout=10)
+        self.pool.shutdown(wait=False)
 
 
 class _null_ctx:
