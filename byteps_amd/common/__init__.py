@@ -54,6 +54,12 @@ _state = _State()
 
 
 def _default_backend() -> str:
+    # BPS_BACKEND=gloo lets a multi-process run share one GPU for compute
+    # while collectives go over gloo (RCCL refuses two ranks on one
+    # device) — the single-box dry-run vehicle for the 8-GPU launch shape
+    forced = os.environ.get("BPS_BACKEND", "").lower()
+    if forced in ("gloo", "nccl"):
+        return forced
     return "nccl" if torch.cuda.is_available() else "gloo"
 
 

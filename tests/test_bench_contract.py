@@ -51,3 +51,38 @@ def test_bench_torchrun_two_ranks():
     assert out["config"]["parallelism"] == "dp2"
     # whole-job aggregate: 2 ranks → global batch = 2 × per-rank
     assert out["config"]["global_batch"] == 128
+
+
+def test_bench_torchrun_four_ranks():
+    """The driver's SCALE run launches N=4 the same way — 4-rank gloo
+    must work end-to-end before it meets RCCL."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()), "bench.py", "--gpus", "4",
+         "--model", "mlp", "--steps", "3", "--warmup", "1",
+         "--device", "cpu"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + "\n" + r.stderr
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    out = _check_json(lines[0], 4)
+    assert out["config"]["parallelism"] == "dp4"
+
+
+def test_bench_torchrun_resnet50_two_ranks():
+    """The flagship model itself (not just the MLP) through the torchrun
+    shape: channels_last buckets + DDP broadcast + engine collectives."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()), "bench.py", "--gpus", "2",
+         "--model", "resnet50", "--batch-size", "1", "--steps", "1",
+         "--warmup", "0", "--device", "cpu"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + "\n" + r.stderr
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    out = _check_json(lines[0], 2)
+    assert out["config"]["model"] == "resnet50"
