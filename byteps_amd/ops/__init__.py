@@ -70,6 +70,52 @@ def scale_(t: torch.Tensor, alpha: float) -> torch.Tensor:
     return t
 
 
+def cast_scale_(dst: torch.Tensor, src: torch.Tensor,
+                alpha: float) -> torch.Tensor:
+    """dst = float(src) * alpha in one fused pass — the averaging divide
+    folded into the reduced-precision wire cast-back."""
+    if dst.is_cuda:
+        _check_gpu(dst), _check_gpu(src)
+        assert dst.numel() == src.numel()
+        core().cast_scale(dst.data_ptr(), src.data_ptr(), dst.numel(), alpha,
+                          _code(src), _code(dst), _stream(dst))
+    else:
+        torch.mul(src.reshape(-1).to(dst.dtype), alpha,
+                  out=dst.reshape(-1))
+    return dst
+
+
+def build_cast_scale_desc(dsts, srcs):
+    """Pack (dst_ptr, src_ptr, vec_prefix) descriptors for
+    :func:`cast_scale_many` into one device int64 tensor.  Every segment's
+    numel must divide the vector width (engine buckets are lcm(64, world)
+    aligned).  Returns (desc_tensor, total_vec, vec)."""
+    n = len(dsts)
+    assert n > 0 and len(srcs) == n
+    vec = 8 if (dsts[0].element_size() == 2 and srcs[0].element_size() == 2) \
+        else 4
+    arr = torch.empty(3 * n + 1, dtype=torch.int64)
+    pref = 0
+    for i, (d, s) in enumerate(zip(dsts, srcs)):
+        assert d.numel() == s.numel() and d.numel() % vec == 0
+        arr[i] = d.data_ptr()
+        arr[n + i] = s.data_ptr()
+        arr[2 * n + i] = pref
+        pref += d.numel() // vec
+    arr[3 * n] = pref
+    return arr.to(dsts[0].device), pref, vec
+
+
+def cast_scale_many_(desc: torch.Tensor, nseg: int, total_vec: int,
+                     alpha: float, src_dtype: torch.dtype,
+                     dst_dtype: torch.dtype, device) -> None:
+    """One launch over all buckets: dst_i = float(src_i) * alpha."""
+    core().cast_scale_many(
+        desc.data_ptr(), nseg, total_vec, alpha,
+        _DTYPE_CODE[src_dtype], _DTYPE_CODE[dst_dtype],
+        torch.cuda.current_stream(device).cuda_stream)
+
+
 def axpy_(y: torch.Tensor, x: torch.Tensor, alpha: float = 1.0) -> torch.Tensor:
     """In-place y += alpha*x (fp32 accumulate for bf16)."""
     if y.is_cuda:

@@ -19,6 +19,11 @@ namespace py = pybind11;
 // -- HIP kernel launchers (kernels.hip / compress.hip) ----------------------
 extern "C" {
 int bps_scale(void* x, int64_t n, float alpha, int dtype, void* stream);
+int bps_cast_scale_many(const void* desc_dev, int nseg, int64_t total_vec,
+                        float alpha, int src_dtype, int dst_dtype,
+                        void* stream);
+int bps_cast_scale(void* dst, const void* src, int64_t n, float alpha,
+                   int src_dtype, int dst_dtype, void* stream);
 int bps_axpy(void* y, const void* x, int64_t n, float alpha, int dtype,
              void* stream);
 int bps_nesterov(void* g, void* m, int64_t n, float mu, int dtype,
@@ -137,6 +142,18 @@ PYBIND11_MODULE(_core, m) {
   m.def("axpy",
         [](uintptr_t y, uintptr_t x, int64_t n, float a, int dt, uintptr_t s) {
           check(bps_axpy(P(y), CP(x), n, a, dt, P(s)), "bps_axpy");
+        });
+  m.def("cast_scale_many",
+        [](uintptr_t desc, int nseg, int64_t total_vec, float a, int sdt,
+           int ddt, uintptr_t s) {
+          check(bps_cast_scale_many(CP(desc), nseg, total_vec, a, sdt, ddt,
+                                    P(s)), "bps_cast_scale_many");
+        });
+  m.def("cast_scale",
+        [](uintptr_t dst, uintptr_t src, int64_t n, float a, int sdt,
+           int ddt, uintptr_t s) {
+          check(bps_cast_scale(P(dst), CP(src), n, a, sdt, ddt, P(s)),
+                "bps_cast_scale");
         });
   m.def("nesterov",
         [](uintptr_t g, uintptr_t mm, int64_t n, float mu, int dt,

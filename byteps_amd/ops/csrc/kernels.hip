@@ -178,6 +178,79 @@ __global__ void norm_f32_kernel(const float* __restrict__ x, int64_t n,
   }
 }
 
+// ---------------------------------------------------------------------------
+// cast_scale: dst = (DST)(float(src) * a) — the post-collective averaging
+// divide fused with the reduced-precision wire cast-back, over MANY
+// buckets in ONE launch (replaces per-bucket copy_ + _foreach_div_ on the
+// default hot path; the reference divided per-tensor on the framework
+// side, torch/ops.cc:78-91).
+//
+// desc layout (int64, device memory): [dst_ptr x nseg][src_ptr x nseg]
+// [vec_prefix x nseg+1].  Segment lengths are multiples of VEC (bucket
+// alignment is lcm(64, world) elements), so there are no tails and every
+// access is a full 16/8-byte vector.
+// ---------------------------------------------------------------------------
+
+using half_t = __half;
+
+__device__ inline float to_f(float v) { return v; }
+__device__ inline float to_f(bf16 v) { return __bfloat162float(v); }
+__device__ inline float to_f(half_t v) { return __half2float(v); }
+__device__ inline void from_f(float v, float* d) { *d = v; }
+__device__ inline void from_f(float v, bf16* d) { *d = __float2bfloat16(v); }
+__device__ inline void from_f(float v, half_t* d) { *d = __float2half(v); }
+
+template <typename SRC, typename DST, int VEC>
+__global__ void cast_scale_many_kernel(const int64_t* __restrict__ desc,
+                                       int nseg, int64_t total_vec,
+                                       float a) {
+  const int64_t* dstp = desc;
+  const int64_t* srcp = desc + nseg;
+  const int64_t* pref = desc + 2 * nseg;  // nseg+1 entries, vec units
+  int64_t i = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * BLOCK;
+  for (; i < total_vec; i += stride) {
+    int lo = 0, hi = nseg - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (pref[mid] <= i) lo = mid; else hi = mid - 1;
+    }
+    int64_t off = (i - pref[lo]) * VEC;
+    const SRC* s = reinterpret_cast<const SRC*>(srcp[lo]) + off;
+    DST* d = reinterpret_cast<DST*>(dstp[lo]) + off;
+    struct __attribute__((aligned(8))) SV { SRC v[VEC]; };
+    struct __attribute__((aligned(8))) DV { DST v[VEC]; };
+    SV sv = *reinterpret_cast<const SV*>(s);
+    DV dv;
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) from_f(to_f(sv.v[j]) * a, &dv.v[j]);
+    *reinterpret_cast<DV*>(d) = dv;
+  }
+}
+
+// single-segment variant (cross-barrier per-bucket path): no descriptor
+// indirection, tail handled scalar
+template <typename SRC, typename DST>
+__global__ void cast_scale_kernel(DST* __restrict__ d,
+                                  const SRC* __restrict__ s, int64_t n,
+                                  float a) {
+  int64_t i0 = ((int64_t)blockIdx.x * BLOCK + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * BLOCK * 4;
+  for (int64_t i = i0; i + 3 < n; i += stride) {
+    struct __attribute__((aligned(8))) SV { SRC v[4]; };
+    struct __attribute__((aligned(8))) DV { DST v[4]; };
+    SV sv = *reinterpret_cast<const SV*>(s + i);
+    DV dv;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) from_f(to_f(sv.v[j]) * a, &dv.v[j]);
+    *reinterpret_cast<DV*>(d + i) = dv;
+  }
+  int64_t tail_start = (n / 4) * 4;
+  int64_t t = tail_start + blockIdx.x * BLOCK + threadIdx.x;
+  if (t < n && blockIdx.x * BLOCK + threadIdx.x < 4)
+    from_f(to_f(s[t]) * a, &d[t]);
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -221,6 +294,44 @@ int bps_nesterov(void* g, void* m, int64_t n, float mu, int dtype,
   if (dtype != 0) return -1;
   hipLaunchKernelGGL(nesterov_f32_kernel, dim3(grid_for(n)), dim3(BLOCK), 0,
                      STREAM, (float*)g, (float*)m, n, mu);
+  return (int)hipGetLastError();
+}
+
+// dst/src dtype codes: 0=f32, 1=f16, 2=bf16.  total_vec = total elements
+// divided by the vector width (4 for mixed/f32 pairs, 8 for 16-bit→16-bit).
+int bps_cast_scale_many(const void* desc_dev, int nseg, int64_t total_vec,
+                        float alpha, int src_dtype, int dst_dtype,
+                        void* stream) {
+  int g = grid_for(total_vec);
+  const int64_t* d = (const int64_t*)desc_dev;
+#define LAUNCH(S, D, V)                                                     \
+  hipLaunchKernelGGL((cast_scale_many_kernel<S, D, V>), dim3(g),            \
+                     dim3(BLOCK), 0, STREAM, d, nseg, total_vec, alpha)
+  if (src_dtype == 0 && dst_dtype == 0) LAUNCH(float, float, 4);
+  else if (src_dtype == 2 && dst_dtype == 0) LAUNCH(bf16, float, 4);
+  else if (src_dtype == 1 && dst_dtype == 0) LAUNCH(half_t, float, 4);
+  else if (src_dtype == 2 && dst_dtype == 2) LAUNCH(bf16, bf16, 8);
+  else if (src_dtype == 1 && dst_dtype == 1) LAUNCH(half_t, half_t, 8);
+  else if (src_dtype == 0 && dst_dtype == 2) LAUNCH(float, bf16, 4);
+  else return -1;
+#undef LAUNCH
+  return (int)hipGetLastError();
+}
+
+int bps_cast_scale(void* dst, const void* src, int64_t n, float alpha,
+                   int src_dtype, int dst_dtype, void* stream) {
+  int g = grid_for((n + 3) / 4);
+#define LAUNCH(S, D)                                                        \
+  hipLaunchKernelGGL((cast_scale_kernel<S, D>), dim3(g), dim3(BLOCK), 0,    \
+                     STREAM, (D*)dst, (const S*)src, n, alpha)
+  if (src_dtype == 0 && dst_dtype == 0) LAUNCH(float, float);
+  else if (src_dtype == 2 && dst_dtype == 0) LAUNCH(bf16, float);
+  else if (src_dtype == 1 && dst_dtype == 0) LAUNCH(half_t, float);
+  else if (src_dtype == 2 && dst_dtype == 2) LAUNCH(bf16, bf16);
+  else if (src_dtype == 1 && dst_dtype == 1) LAUNCH(half_t, half_t);
+  else if (src_dtype == 0 && dst_dtype == 2) LAUNCH(float, bf16);
+  else return -1;
+#undef LAUNCH
   return (int)hipGetLastError();
 }
 

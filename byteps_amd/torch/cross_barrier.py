@@ -140,14 +140,28 @@ class CrossBarrier:
 
     def _finish_bucket(self, bucket: Bucket) -> None:
         eng = self._engine
+        wire_src = None
         if bucket.ps_ticket is not None:
             eng._ps.wait(bucket.ps_ticket)
         elif bucket.work is not None:
             bucket.work.wait()
             if eng.comm_dtype is not None:
-                bucket.buffer.copy_(eng._wire_scratch[bucket.plan.index])
-        if eng.average and eng.world > 1 and not eng.prescale:
-            bucket.buffer.div_(eng.world)
+                wire_src = eng._wire_scratch[bucket.plan.index]
+        divide = eng.average and eng.world > 1 and not eng.prescale
+        if wire_src is not None or divide:
+            alpha = (1.0 / eng.world) if divide else 1.0
+            from .. import ops as _ops
+            if bucket.buffer.is_cuda and _ops.have_core():
+                # fused cast-back + averaging divide, one CDNA4 kernel
+                _ops.cast_scale_(
+                    bucket.buffer,
+                    wire_src if wire_src is not None else bucket.buffer,
+                    alpha)
+            else:
+                if wire_src is not None:
+                    bucket.buffer.copy_(wire_src)
+                if divide:
+                    bucket.buffer.div_(eng.world)
         opt = self._bucket_opts.get(bucket.plan.index)
         if opt is not None:
             self._refresh_hyper(opt)

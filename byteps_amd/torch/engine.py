@@ -258,6 +258,12 @@ class GradEngine:
         if self.comm_dtype is not None and dt == self.comm_dtype:
             self.comm_dtype = None
         self._wire_scratch: Dict[int, torch.Tensor] = {}
+        if self.comm_dtype is not None:
+            # eager: the fused cast+scale descriptor needs stable pointers
+            for b in self.buckets:
+                self._wire_scratch[b.plan.index] = torch.empty_like(
+                    b.buffer, dtype=self.comm_dtype)
+        self._fused_desc = None     # (desc dev tensor, total_vec, vec)
 
         self._ps = None
         if C._state.ps_enabled:
@@ -419,16 +425,17 @@ class GradEngine:
         if not self._sync_enabled:
             return
         self.flush()
+        needs_wire = False
         for b in self.buckets:
             if b.ps_ticket is not None:
                 self._ps.wait(b.ps_ticket)
             elif b.work is not None:
                 b.work.wait()
                 if self.comm_dtype is not None:
-                    b.buffer.copy_(self._wire_scratch[b.plan.index])
-        if self.average and self.world > 1 and not self.prescale:
-            torch._foreach_div_([b.buffer for b in self.buckets],
-                                float(self.world))
+                    needs_wire = True
+        divide = self.average and self.world > 1 and not self.prescale
+        if needs_wire or divide:
+            self._apply_wire_and_average(needs_wire, divide)
         if self._split_params:
             # split params read back their reduced (averaged) grads —
             # unconditionally: with average=False the SUM must still
@@ -453,6 +460,32 @@ class GradEngine:
         self._ready_params = 0
         self._inflight_bytes = 0
         self._step += 1
+
+    def _apply_wire_and_average(self, needs_wire: bool,
+                                divide: bool) -> None:
+        """Post-collective epilogue: the averaging divide fused with the
+        reduced-precision wire cast-back — ONE hand-written CDNA4 kernel
+        launch over all buckets on GPU (bps_cast_scale_many, kernels.hip;
+        replaces per-bucket copy_ + torch._foreach_div_; the reference
+        divided per-tensor on the framework side, torch/ops.cc:78-91)."""
+        alpha = (1.0 / self.world) if divide else 1.0
+        bufs = [b.buffer for b in self.buckets]
+        from .. import ops as _ops
+        if bufs[0].is_cuda and _ops.have_core():
+            if self._fused_desc is None:
+                srcs = [self._wire_scratch[b.plan.index]
+                        for b in self.buckets] if needs_wire else bufs
+                self._fused_desc = _ops.build_cast_scale_desc(bufs, srcs)
+            desc, total_vec, _vec = self._fused_desc
+            src_dt = self.comm_dtype if needs_wire else bufs[0].dtype
+            _ops.cast_scale_many_(desc, len(bufs), total_vec, alpha,
+                                  src_dt, bufs[0].dtype, bufs[0].device)
+        else:
+            if needs_wire:
+                for b in self.buckets:
+                    b.buffer.copy_(self._wire_scratch[b.plan.index])
+            if divide:
+                torch._foreach_div_(bufs, float(self.world))
 
     def detach(self) -> None:
         for h in self._hook_handles:

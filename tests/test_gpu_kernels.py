@@ -173,3 +173,62 @@ def test_fp8_gpu_matches_cpu():
     rel = (out - xg).abs() / (xg.abs() + 1e-9)
     normal = xg.abs() > amax_g / 448 * 2 ** -6
     assert rel[normal].max().item() < 1 / 16 + 1e-3
+
+
+# -- fused cast+scale (hot-path averaging epilogue) --------------------------
+
+@pytest.mark.parametrize("src_dt,dst_dt", [
+    (torch.float32, torch.float32),
+    (torch.bfloat16, torch.float32),
+    (torch.float16, torch.float32),
+    (torch.bfloat16, torch.bfloat16),
+])
+def test_cast_scale_single(src_dt, dst_dt):
+    n = (1 << 18) + 64
+    src = _cuda(n, 7, src_dt)
+    dst = torch.empty(n, dtype=dst_dt, device="cuda")
+    ref = (src.float() * 0.125).to(dst_dt).float()
+    K.cast_scale_(dst, src, 0.125)
+    torch.cuda.synchronize()
+    assert torch.equal(dst.float(), ref)
+
+
+@pytest.mark.parametrize("src_dt,dst_dt", [
+    (torch.float32, torch.float32),
+    (torch.bfloat16, torch.float32),
+    (torch.bfloat16, torch.bfloat16),
+])
+def test_cast_scale_many(src_dt, dst_dt):
+    """Multi-bucket descriptor kernel — the exact engine epilogue shape
+    (segment lengths are multiples of 64, as bucket alignment guarantees)."""
+    sizes = [64, 4096, 1 << 16, 192, (1 << 18)]
+    srcs = [_cuda(n, 11 + i, src_dt) for i, n in enumerate(sizes)]
+    dsts = [torch.empty(n, dtype=dst_dt, device="cuda") for n in sizes]
+    refs = [(s.float() * 0.25).to(dst_dt).float() for s in srcs]
+    desc, total_vec, _vec = K.build_cast_scale_desc(dsts, srcs)
+    K.cast_scale_many_(desc, len(sizes), total_vec, 0.25, src_dt, dst_dt,
+                       dsts[0].device)
+    torch.cuda.synchronize()
+    for d, r in zip(dsts, refs):
+        assert torch.equal(d.float(), r)
+
+
+def test_engine_epilogue_uses_fused_kernel():
+    """GradEngine at world=1 never averages, so drive the epilogue
+    directly: buffers divided in-place by the descriptor kernel must match
+    _foreach_div_."""
+    import byteps_amd.torch as bps
+    from byteps_amd.torch.engine import GradEngine
+    bps.init()
+    m = torch.nn.Sequential(torch.nn.Linear(128, 256),
+                            torch.nn.Linear(256, 32)).cuda()
+    eng = GradEngine(list(m.named_parameters()))
+    for b in eng.buckets:
+        b.buffer.normal_()
+    refs = [b.buffer.float() / 4.0 for b in eng.buckets]
+    eng.world = 4          # pretend world for the divide
+    eng._apply_wire_and_average(needs_wire=False, divide=True)
+    torch.cuda.synchronize()
+    for b, r in zip(eng.buckets, refs):
+        assert torch.allclose(b.buffer.float(), r, rtol=1e-6, atol=1e-7)
+    bps.shutdown()
