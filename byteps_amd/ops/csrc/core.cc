@@ -95,6 +95,8 @@ int bps_cpu_sum2(void* dst, const void* src1, const void* src2, int64_t n,
                  float alpha, int dtype);
 int bps_cpu_copy(void* dst, const void* src, int64_t nbytes);
 int bps_cpu_scale(void* x, int64_t n, float alpha, int dtype);
+int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
+                        float* val);
 int bps_cpu_onebit_compress(const float* x, int64_t n, uint64_t* bits,
                             float* scale_sum);
 int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
@@ -328,6 +330,12 @@ PYBIND11_MODULE(_core, m) {
   m.def("cpu_scale", [](uintptr_t x, int64_t n, float a, int dt) {
     check(bps_cpu_scale(P(x), n, a, dt), "bps_cpu_scale");
   }, py::call_guard<py::gil_scoped_release>());
+  m.def("cpu_topk_select",
+        [](uintptr_t x, int64_t n, int64_t k, uintptr_t idx, uintptr_t val) {
+          check(bps_cpu_topk_select((const float*)CP(x), n, k,
+                                    (int32_t*)P(idx), (float*)P(val)),
+                "bps_cpu_topk_select");
+        }, py::call_guard<py::gil_scoped_release>());
   m.def("cpu_onebit_compress", [](uintptr_t x, int64_t n, uintptr_t bits) {
     float sc = 0.0f;
     check(bps_cpu_onebit_compress((const float*)P(x), n, (uint64_t*)P(bits),

@@ -10,6 +10,7 @@
 #include <cmath>
 #include <cstdint>
 #include <cstring>
+#include <vector>
 
 #ifdef _OPENMP
 #include <omp.h>
@@ -223,6 +224,72 @@ int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
 #pragma omp parallel for
   for (int64_t i = 0; i < n; ++i)
     out[i] = fp8_e4m3_decode(code[i]) * inv;
+  return 0;
+}
+
+// Parallel top-k by |x|: per-thread min-heaps over disjoint ranges, then
+// one partial_sort over the k*T candidates (k log k·T) — replaces the
+// single-threaded O(n log k) partial_sort over all n that made topk the
+// slowest server codec (reference used a serial heap
+// too, impl/topk.cc:43-78).  Output pairs are ordered by descending |x|.
+int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
+                        float* val) {
+  if (k <= 0) return -1;
+  if (k > n) k = n;
+  int nt = 1;
+#ifdef _OPENMP
+  nt = omp_get_max_threads();
+#endif
+  if ((int64_t)nt * 4 > n / (k > 0 ? k : 1) + 1) nt = 1;  // tiny inputs
+  std::vector<int32_t> cand_idx((size_t)nt * k);
+  std::vector<int64_t> cand_cnt(nt, 0);
+#pragma omp parallel num_threads(nt)
+  {
+#ifdef _OPENMP
+    int t = omp_get_thread_num();
+#else
+    int t = 0;
+#endif
+    int64_t lo = n * t / nt, hi = n * (t + 1) / nt;
+    int32_t* heap = cand_idx.data() + (size_t)t * k;
+    int64_t cnt = 0;
+    auto less = [&](int32_t a, int32_t b) {  // min-heap on |x|
+      return std::fabs(x[a]) > std::fabs(x[b]);
+    };
+    for (int64_t i = lo; i < hi; ++i) {
+      if (cnt < k) {
+        heap[cnt++] = (int32_t)i;
+        if (cnt == k) std::make_heap(heap, heap + k, less);
+      } else if (std::fabs(x[i]) > std::fabs(x[heap[0]])) {
+        std::pop_heap(heap, heap + k, less);
+        heap[k - 1] = (int32_t)i;
+        std::push_heap(heap, heap + k, less);
+      }
+    }
+    cand_cnt[t] = cnt;
+  }
+  // compact candidates (ranges shorter than k contribute fewer)
+  std::vector<int32_t> all;
+  all.reserve((size_t)nt * k);
+  for (int t = 0; t < nt; ++t) {
+    const int32_t* c = cand_idx.data() + (size_t)t * k;
+    all.insert(all.end(), c, c + cand_cnt[t]);
+  }
+  if ((int64_t)all.size() > k) {
+    std::partial_sort(all.begin(), all.begin() + k, all.end(),
+                      [&](int32_t a, int32_t b) {
+                        return std::fabs(x[a]) > std::fabs(x[b]);
+                      });
+    all.resize(k);
+  } else {
+    std::sort(all.begin(), all.end(), [&](int32_t a, int32_t b) {
+      return std::fabs(x[a]) > std::fabs(x[b]);
+    });
+  }
+  for (int64_t j = 0; j < (int64_t)all.size(); ++j) {
+    idx[j] = all[j];
+    val[j] = x[all[j]];
+  }
   return 0;
 }
 

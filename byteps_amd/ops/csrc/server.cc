@@ -66,6 +66,8 @@ int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
                                int natural, float norm, int8_t* code);
 int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
                                  int natural, float norm, float* out);
+int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
+                        float* val);
 int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
                          uint8_t* code);
 int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
@@ -592,21 +594,13 @@ class Server {
         break;
       }
       case kTopk: {
+        // k for topk is carried in `levels` at init; parallel per-thread
+        // heap select (cpu_reducer.cc) — was a serial O(n log k)
+        // partial_sort, the slowest codec on the merge path
         int64_t k = std::max<int64_t>(1, std::min<int64_t>(n, (int64_t)ks->levels));
-        // k for topk is carried in `levels` at init
-        std::vector<int32_t> order(n);
-        for (int64_t i = 0; i < n; ++i) order[i] = (int32_t)i;
-        std::partial_sort(order.begin(), order.begin() + k, order.end(),
-                          [acc](int32_t a, int32_t b) {
-                            return std::fabs(acc[a]) > std::fabs(acc[b]);
-                          });
         ks->reply.resize(k * 8);
-        int32_t* idx = (int32_t*)ks->reply.data();
-        float* val = (float*)(ks->reply.data() + k * 4);
-        for (int64_t j = 0; j < k; ++j) {
-          idx[j] = order[j];
-          val[j] = acc[order[j]];
-        }
+        bps_cpu_topk_select(acc, n, k, (int32_t*)ks->reply.data(),
+                            (float*)(ks->reply.data() + k * 4));
         break;
       }
       case kRandomk: {

@@ -125,3 +125,22 @@ def test_telemetry_window_decay():
     m.record(1_000_000)
     _, mbps = m.speed()
     assert mbps >= 0.0
+
+
+def test_cpu_topk_select_parallel():
+    """Parallel per-thread-heap top-k matches torch.topk's selection set
+    (server reply path, replaces the serial partial_sort)."""
+    import torch
+    from byteps_amd.ops import core
+    c = core()
+    for n, k in [(1000, 10), (1 << 16, 256), (5000, 5000), (64, 128)]:
+        x = torch.randn(n)
+        kk = min(k, n)
+        idx = torch.empty(kk, dtype=torch.int32)
+        val = torch.empty(kk)
+        c.cpu_topk_select(x.data_ptr(), n, k, idx.data_ptr(), val.data_ptr())
+        _rv, ref_i = torch.topk(x.abs(), kk)
+        assert set(idx.tolist()) == set(ref_i.tolist())
+        assert torch.allclose(val, x[idx.long()])
+        a = x[idx.long()].abs()
+        assert bool((a[:-1] >= a[1:]).all())
