@@ -11,11 +11,16 @@
 // common/core_loops.cc:185).
 
 #include <arpa/inet.h>
+#include <fcntl.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <sys/mman.h>
 #include <sys/socket.h>
+#include <sys/stat.h>
 #include <sys/uio.h>
 #include <unistd.h>
+
+#include <hip/hip_runtime_api.h>
 
 #include <atomic>
 #include <condition_variable>
@@ -94,6 +99,8 @@ struct Request {
   const void* send_payload = nullptr;  // for push
   void* recv_buf = nullptr;            // for pull
   size_t recv_cap = 0;
+  bool use_ext = false;                // IPC: body is the 16-byte IpcExt
+  IpcExt ext{0, 0};
   // completion
   std::mutex mu;
   std::condition_variable cv;
@@ -103,13 +110,117 @@ struct Request {
   std::string error;
 };
 
+// One shm region shared with a colocated server (the client's staging
+// arena IS the transport: push payloads are read by the server in place,
+// pull replies are written by the server straight into the staging the
+// GPU H2D reads from — zero socket copies for gradient bytes).
+struct Region {
+  std::string name;
+  char* base = nullptr;
+  uint64_t size = 0;
+  uint64_t used = 0;
+  bool hip_registered = false;
+};
+
 class ServerConn {
  public:
   ServerConn(const std::string& host, int port, uint32_t rank)
       : rank_(rank) {
     fd_ = connect_to(host, port);
+    const char* e = getenv("BPS_ENABLE_IPC");
+    ipc_enabled_ = !(e && (e[0] == '0' || e[0] == 'n' || e[0] == 'N'));
+    const char* rm = getenv("BPS_IPC_REGION_MB");
+    default_region_ = (rm ? std::max(1L, atol(rm)) : 64L) * (1ULL << 20);
     sender_ = std::thread([this] { this->send_loop(); });
     receiver_ = std::thread([this] { this->recv_loop(); });
+  }
+
+  // Allocate staging bytes inside a shm region shared with the server.
+  // Returns the mapped address, or 0 when IPC is unavailable (remote
+  // server / disabled) — caller falls back to private pinned staging.
+  uintptr_t ipc_alloc(uint64_t nbytes) {
+    if (!ipc_enabled_ || closed_) return 0;
+    std::lock_guard<std::mutex> lk(regions_mu_);
+    if (!ipc_enabled_) return 0;
+    nbytes = (nbytes + 255) & ~255ULL;
+    for (auto& r : regions_) {
+      if (r.size - r.used >= nbytes) {
+        uintptr_t p = (uintptr_t)(r.base + r.used);
+        r.used += nbytes;
+        return p;
+      }
+    }
+    Region r;
+    r.size = std::max<uint64_t>(default_region_, nbytes);
+    static std::atomic<uint32_t> ctr{0};
+    char nm[96];
+    snprintf(nm, sizeof(nm), "/bpsamd-%d-%u-%u", (int)getpid(), rank_,
+             ctr.fetch_add(1));
+    r.name = nm;
+    int fd = shm_open(nm, O_CREAT | O_EXCL | O_RDWR, 0600);
+    if (fd < 0 || ftruncate(fd, (off_t)r.size) != 0) {
+      if (fd >= 0) { ::close(fd); shm_unlink(nm); }
+      ipc_enabled_ = false;
+      return 0;
+    }
+    r.base = (char*)mmap(nullptr, r.size, PROT_READ | PROT_WRITE,
+                         MAP_SHARED, fd, 0);
+    ::close(fd);
+    if (r.base == MAP_FAILED) {
+      shm_unlink(nm);
+      ipc_enabled_ = false;
+      return 0;
+    }
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) == hipSuccess && ndev > 0) {
+      // pin for fast async D2H/H2D; ignore failure (stays pageable)
+      r.hip_registered =
+          hipHostRegister(r.base, r.size, hipHostRegisterPortable) ==
+          hipSuccess;
+      (void)hipGetLastError();
+    }
+    // hello handshake: the server shm_opens the name — success proves
+    // colocation (a remote server cannot see this host's /dev/shm)
+    uint32_t region_id = (uint32_t)regions_.size();
+    auto req = submit(kIpcHello, region_id, r.name.c_str(),
+                      r.name.size() + 1, nullptr, 0, 0, r.size);
+    uint64_t aux = ~0ULL;
+    {
+      std::unique_lock<std::mutex> rlk(req->mu);
+      req->cv.wait(rlk, [&] { return req->done; });
+      if (req->error.empty()) aux = req->reply_aux;
+    }
+    shm_unlink(nm);  // server holds its own mapping (or failed)
+    if (aux == ~0ULL) {
+      if (r.hip_registered) (void)hipHostUnregister(r.base);
+      munmap(r.base, r.size);
+      ipc_enabled_ = false;
+      return 0;
+    }
+    uintptr_t p = (uintptr_t)r.base;
+    r.used = nbytes;
+    regions_.push_back(r);
+    return p;
+  }
+
+  bool ipc_active() {
+    std::lock_guard<std::mutex> lk(regions_mu_);
+    return ipc_enabled_ && !regions_.empty();
+  }
+
+  // locate a pointer range inside a shared region → wire locator
+  bool locate(const void* p, uint64_t len, uint64_t* loc) {
+    if (!p) return false;
+    std::lock_guard<std::mutex> lk(regions_mu_);
+    const char* c = (const char*)p;
+    for (size_t i = 0; i < regions_.size(); ++i) {
+      const Region& r = regions_[i];
+      if (c >= r.base && c + len <= r.base + r.size) {
+        *loc = make_locator((uint32_t)i, (uint64_t)(c - r.base));
+        return true;
+      }
+    }
+    return false;
   }
 
   ~ServerConn() { close(); }
@@ -126,9 +237,18 @@ class ServerConn {
     if (receiver_.joinable()) receiver_.join();
     ::close(fd_);
     // fail anything still outstanding
-    std::lock_guard<std::mutex> lk(inflight_mu_);
-    for (auto& kv : inflight_) complete(kv.second, 0, 0, "connection closed");
-    inflight_.clear();
+    {
+      std::lock_guard<std::mutex> lk(inflight_mu_);
+      for (auto& kv : inflight_)
+        complete(kv.second, 0, 0, "connection closed");
+      inflight_.clear();
+    }
+    std::lock_guard<std::mutex> rlk(regions_mu_);
+    for (auto& r : regions_) {
+      if (r.hip_registered) (void)hipHostUnregister(r.base);
+      munmap(r.base, r.size);
+    }
+    regions_.clear();
   }
 
   std::shared_ptr<Request> submit(uint32_t op, uint64_t key,
@@ -136,11 +256,26 @@ class ServerConn {
                                   void* recv_buf, size_t recv_cap,
                                   uint32_t cmd, uint64_t aux) {
     auto req = std::make_shared<Request>();
-    req->hdr = MsgHeader{kMagic, op, key, (op == kPush || op == kInit) ? len : 0,
+    req->hdr = MsgHeader{kMagic, op, key,
+                         (op == kPush || op == kInit || op == kIpcHello)
+                             ? len : 0,
                          aux, rank_, cmd, seq_.fetch_add(1)};
     req->send_payload = payload;
     req->recv_buf = recv_buf;
     req->recv_cap = recv_cap;
+    // colocated fast path: payload / receive buffer inside a shared
+    // region → ship a 16-byte locator instead of the bytes
+    uint64_t loc;
+    if (op == kPush && len > 0 && locate(payload, len, &loc)) {
+      req->use_ext = true;
+      req->ext = IpcExt{loc, 0};
+      req->hdr.cmd |= kCmdIpcPayload;
+    } else if (op == kPull && recv_buf &&
+               locate(recv_buf, recv_cap, &loc)) {
+      req->use_ext = true;
+      req->ext = IpcExt{loc, recv_cap};
+      req->hdr.cmd |= kCmdIpcPayload;
+    }
     {
       std::lock_guard<std::mutex> lk(inflight_mu_);
       inflight_[req->hdr.seq] = req;
@@ -166,7 +301,12 @@ class ServerConn {
         queue_.pop_front();
       }
       try {
-        if (req->hdr.len > 0 && req->send_payload) {
+        if (req->use_ext) {
+          char frame[sizeof(MsgHeader) + sizeof(IpcExt)];
+          std::memcpy(frame, &req->hdr, sizeof(MsgHeader));
+          std::memcpy(frame + sizeof(MsgHeader), &req->ext, sizeof(IpcExt));
+          write_all(fd_, frame, sizeof(frame));
+        } else if (req->hdr.len > 0 && req->send_payload) {
           iovec iov[2];
           iov[0].iov_base = &req->hdr;
           iov[0].iov_len = sizeof(MsgHeader);
@@ -214,7 +354,12 @@ class ServerConn {
           inflight_.erase(it);
         }
       }
-      if (h.len > 0) {
+      if (h.cmd & kCmdIpcPayload) {
+        // reply body is the 16-byte locator echo; the payload is already
+        // in the shared region (= the caller's receive buffer)
+        IpcExt ext;
+        if (!read_all(fd_, &ext, sizeof(ext))) return;
+      } else if (h.len > 0) {
         void* dst = nullptr;
         if (req && req->recv_buf && h.len <= req->recv_cap) {
           dst = req->recv_buf;  // zero-copy into caller's pinned buffer
@@ -254,6 +399,10 @@ class ServerConn {
   uint32_t rank_;
   std::atomic<uint64_t> seq_{1};
   std::atomic<bool> closed_{false};
+  bool ipc_enabled_ = false;
+  uint64_t default_region_ = 64ULL << 20;
+  std::mutex regions_mu_;
+  std::vector<Region> regions_;
   std::mutex q_mu_;
   std::condition_variable q_cv_;
   std::deque<std::shared_ptr<Request>> queue_;
@@ -320,6 +469,14 @@ class KVClient {
     for (auto& c : conns_) c->close();
   }
 
+  // colocated IPC staging: returns the shm-backed address (0 = fall back
+  // to private pinned staging)
+  uintptr_t ipc_alloc(int server, uint64_t nbytes) {
+    return conns_.at(server)->ipc_alloc(nbytes);
+  }
+
+  bool ipc_active(int server) { return conns_.at(server)->ipc_active(); }
+
   int num_servers() const { return (int)conns_.size(); }
 
  private:
@@ -348,6 +505,10 @@ void init_kv(py::module_& m) {
       .def("test", &KVClient::test,
            py::call_guard<py::gil_scoped_release>())
       .def("close", &KVClient::close,
+           py::call_guard<py::gil_scoped_release>())
+      .def("ipc_alloc", &KVClient::ipc_alloc,
+           py::call_guard<py::gil_scoped_release>())
+      .def("ipc_active", &KVClient::ipc_active,
            py::call_guard<py::gil_scoped_release>())
       .def_property_readonly("num_servers", &KVClient::num_servers);
 

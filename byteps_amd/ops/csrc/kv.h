@@ -21,7 +21,29 @@ enum MsgOp : uint32_t {
   kBarrier = 7,
   kBarrierReply = 8,
   kShutdown = 9,
+  kIpcHello = 10,   // announce a shm region: payload = shm name,
+                    // aux = size, key = region id (colocated fast path)
+  kIpcHelloReply = 11,
 };
+
+// Colocated IPC fast path: when cmd has kCmdIpcPayload set, the frame
+// body is a fixed 16-byte IpcExt (locator + capacity) instead of the
+// payload — the payload itself lives in a shm region the client
+// announced with kIpcHello, so gradient bytes never cross the socket
+// (reference ps-lite had an IPC van for the same reason,
+// docs/best-practice.md:32).
+constexpr uint32_t kCmdIpcPayload = 1u << 24;
+
+struct IpcExt {
+  uint64_t locator;  // region_id << 48 | byte offset
+  uint64_t cap;      // pull request: receive capacity at locator
+};
+
+inline uint64_t make_locator(uint32_t region, uint64_t off) {
+  return ((uint64_t)region << 48) | (off & ((1ULL << 48) - 1));
+}
+inline uint32_t locator_region(uint64_t l) { return (uint32_t)(l >> 48); }
+inline uint64_t locator_off(uint64_t l) { return l & ((1ULL << 48) - 1); }
 
 // codec ids on the wire
 enum WireCodec : uint32_t {

@@ -20,9 +20,12 @@
 //    immediately (reference server/server.cc:315-319).
 
 #include <arpa/inet.h>
+#include <fcntl.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <sys/mman.h>
 #include <sys/socket.h>
+#include <sys/stat.h>
 #include <unistd.h>
 
 #include <algorithm>
@@ -96,6 +99,42 @@ bool read_all_fd(int fd, void* buf, size_t n) {
 struct Conn {
   int fd;
   std::mutex write_mu;
+  // colocated IPC: shm regions announced by this worker (kIpcHello);
+  // reader thread installs, engine threads look up
+  std::mutex regions_mu;
+  std::unordered_map<uint32_t, std::pair<char*, size_t>> regions;
+
+  ~Conn() {
+    for (auto& kv : regions) munmap(kv.second.first, kv.second.second);
+  }
+
+  char* resolve(uint64_t locator, uint64_t len) {
+    std::lock_guard<std::mutex> lk(regions_mu);
+    auto it = regions.find(bpsamd::locator_region(locator));
+    if (it == regions.end()) return nullptr;
+    uint64_t off = bpsamd::locator_off(locator);
+    if (off + len > it->second.second) return nullptr;  // bounds check
+    return it->second.first + off;
+  }
+
+  // header + 16-byte IpcExt frame (reply for region-backed payloads)
+  void send_ext(const MsgHeader& h, const bpsamd::IpcExt& ext) {
+    char frame[sizeof(MsgHeader) + sizeof(bpsamd::IpcExt)];
+    std::memcpy(frame, &h, sizeof(MsgHeader));
+    std::memcpy(frame + sizeof(MsgHeader), &ext, sizeof(ext));
+    std::lock_guard<std::mutex> lk(write_mu);
+    const char* p = frame;
+    size_t n = sizeof(frame);
+    while (n > 0) {
+      ssize_t w = ::write(fd, p, n);
+      if (w < 0) {
+        if (errno == EINTR) continue;
+        return;
+      }
+      p += w;
+      n -= (size_t)w;
+    }
+  }
 
   void send(const MsgHeader& h, const void* payload) {
     std::lock_guard<std::mutex> lk(write_mu);
@@ -137,6 +176,8 @@ struct InitPayload {
 struct PendingPull {
   std::shared_ptr<Conn> conn;
   MsgHeader hdr;
+  bool ipc = false;
+  IpcExt ext{0, 0};
 };
 
 struct KeyState {
@@ -168,6 +209,11 @@ struct Task {
   MsgHeader hdr;
   std::vector<char> payload;
   KeyState* ks;
+  const char* ipc_payload = nullptr;  // payload lives in the worker's shm
+
+  const char* data() const {
+    return ipc_payload ? ipc_payload : payload.data();
+  }
 };
 
 class Server {
@@ -307,10 +353,29 @@ class Server {
       MsgHeader h;
       if (!read_all_fd(conn->fd, &h, sizeof(h))) return;
       if (h.magic != kMagic) return;
-      std::vector<char> payload(h.len);
-      if (h.len > 0 && !read_all_fd(conn->fd, payload.data(), h.len)) return;
+      const bool ipc = (h.cmd & kCmdIpcPayload) != 0;
+      IpcExt ext{0, 0};
+      std::vector<char> payload;
+      const char* ipc_ptr = nullptr;
+      if (ipc) {
+        // body is the 16-byte locator; payload lives in the worker's shm
+        if (!read_all_fd(conn->fd, &ext, sizeof(ext))) return;
+        if (h.op == kPush) {
+          ipc_ptr = conn->resolve(ext.locator, h.len);
+          if (!ipc_ptr) {
+            reply_err(conn, h);
+            continue;
+          }
+        }
+      } else if (h.len > 0) {
+        payload.resize(h.len);
+        if (!read_all_fd(conn->fd, payload.data(), h.len)) return;
+      }
 
       switch (h.op) {
+        case kIpcHello:
+          handle_ipc_hello(conn, h, payload);
+          break;
         case kInit:
           handle_init(conn, h, payload);
           break;
@@ -325,7 +390,9 @@ class Server {
             ks->push_total++;
           }
           int tid = engine_of(h.key);
-          queues_[tid]->push(Task{conn, h, std::move(payload), ks});
+          Task t{conn, h, std::move(payload), ks};
+          t.ipc_payload = ipc_ptr;
+          queues_[tid]->push(std::move(t));
           break;
         }
         case kPull: {
@@ -334,7 +401,7 @@ class Server {
             reply_err(conn, h);
             break;
           }
-          handle_pull(conn, h, ks);
+          handle_pull(conn, h, ks, ipc, ext);
           break;
         }
         case kBarrier:
@@ -348,10 +415,38 @@ class Server {
     }
   }
 
+  void handle_ipc_hello(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
+                        const std::vector<char>& payload) {
+    MsgHeader r = h;
+    r.op = kIpcHelloReply;
+    r.len = 0;
+    // shm_open succeeding proves colocation: a remote worker's region
+    // name does not exist in this host's /dev/shm
+    char* base = nullptr;
+    if (!payload.empty() && payload.back() == '\0' && h.aux > 0) {
+      int fd = shm_open(payload.data(), O_RDWR, 0600);
+      if (fd >= 0) {
+        base = (char*)mmap(nullptr, h.aux, PROT_READ | PROT_WRITE,
+                           MAP_SHARED, fd, 0);
+        ::close(fd);
+        if (base == MAP_FAILED) base = nullptr;
+      }
+    }
+    if (base) {
+      std::lock_guard<std::mutex> lk(conn->regions_mu);
+      conn->regions[(uint32_t)h.key] = {base, (size_t)h.aux};
+      r.aux = h.key;
+    } else {
+      r.aux = ~0ULL;  // not colocated / cannot map
+    }
+    conn->send(r, nullptr);
+  }
+
   void reply_err(const std::shared_ptr<Conn>& conn, const MsgHeader& req) {
     MsgHeader h = req;
     h.len = 0;
     h.aux = ~0ULL;  // error marker
+    h.cmd &= ~kCmdIpcPayload;  // no ext follows
     h.op = req.op == kPull ? kPullReply
          : req.op == kInit ? kInitReply : kPushReply;
     conn->send(h, nullptr);
@@ -418,32 +513,50 @@ class Server {
   }
 
   void handle_pull(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
-                   KeyState* ks) {
+                   KeyState* ks, bool ipc, const IpcExt& ext) {
     std::unique_lock<std::mutex> lk(ks->mu);
     uint64_t want_version = h.aux;
+    PendingPull p{conn, h, ipc, ext};
     if (ks->async_mode || ks->version >= want_version) {
-      send_pull_reply(conn, h, ks);
+      send_pull_reply(p, ks);
     } else {
-      ks->pending.push_back(PendingPull{conn, h});
+      ks->pending.push_back(std::move(p));
     }
   }
 
-  void send_pull_reply(const std::shared_ptr<Conn>& conn, const MsgHeader& req,
-                       KeyState* ks) {
+  void send_pull_reply(const PendingPull& p, KeyState* ks) {
     // ks->mu held
-    MsgHeader r = req;
+    MsgHeader r = p.hdr;
     r.op = kPullReply;
     r.aux = ks->version;
+    const void* src;
+    uint64_t len;
     if (ks->codec == kRaw) {
-      const std::vector<float>& src =
+      const std::vector<float>& v =
           (ks->async_mode || ks->published.empty()) ? ks->store
                                                     : ks->published;
-      r.len = src.size() * sizeof(float);
-      conn->send(r, src.data());
+      src = v.data();
+      len = v.size() * sizeof(float);
     } else {
-      r.len = ks->reply.size();
-      conn->send(r, ks->reply.data());
+      src = ks->reply.data();
+      len = ks->reply.size();
     }
+    r.len = len;
+    if (p.ipc) {
+      // colocated: write straight into the worker's recv staging —
+      // the only copy on the reply path
+      char* dst = len <= p.ext.cap
+                      ? p.conn->resolve(p.ext.locator, len) : nullptr;
+      if (dst) {
+        std::memcpy(dst, src, len);
+        p.conn->send_ext(r, p.ext);
+        return;
+      }
+      // capacity/region mismatch: fall back to inline (clears the IPC
+      // bit so the client reads the payload from the socket)
+      r.cmd &= ~kCmdIpcPayload;
+    }
+    p.conn->send(r, src);
   }
 
   void engine_loop(int tid) {
@@ -476,7 +589,7 @@ class Server {
 
     switch (codec) {
       case kRaw: {
-        const float* src = (const float*)t.payload.data();
+        const float* src = (const float*)t.data();
         if (first)
           std::memcpy(acc, src, n * sizeof(float));
         else
@@ -486,9 +599,9 @@ class Server {
       case kOnebit: {
         int64_t nwords = (n + 63) >> 6;
         float scale_sum;
-        std::memcpy(&scale_sum, t.payload.data() + nwords * 8, 4);
+        std::memcpy(&scale_sum, t.data() + nwords * 8, 4);
         ks->scratch.resize(n);
-        bps_cpu_onebit_decompress((const uint64_t*)t.payload.data(),
+        bps_cpu_onebit_decompress((const uint64_t*)t.data(),
                                   scale_sum, n, ks->scratch.data());
         if (first)
           std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
@@ -499,8 +612,8 @@ class Server {
       case kTopk:
       case kRandomk: {
         int64_t k = (int64_t)t.hdr.aux & 0xFFFFFFFF;
-        const int32_t* idx = (const int32_t*)t.payload.data();
-        const float* val = (const float*)(t.payload.data() + k * 4);
+        const int32_t* idx = (const int32_t*)t.data();
+        const float* val = (const float*)(t.data() + k * 4);
         if (first) std::memset(acc, 0, n * sizeof(float));
         bps_cpu_sparse_accumulate(idx, val, k, acc);
         break;
@@ -508,8 +621,8 @@ class Server {
       case kDitherLinear:
       case kDitherNatural: {
         float norm;
-        std::memcpy(&norm, t.payload.data(), 4);
-        const int8_t* code = (const int8_t*)(t.payload.data() + 4);
+        std::memcpy(&norm, t.data(), 4);
+        const int8_t* code = (const int8_t*)(t.data() + 4);
         ks->scratch.resize(n);
         bps_cpu_dithering_decompress(code, n, (int)ks->levels,
                                      codec == kDitherNatural, norm,
@@ -522,8 +635,8 @@ class Server {
       }
       case kFp8: {
         float amax;
-        std::memcpy(&amax, t.payload.data(), 4);
-        const uint8_t* code = (const uint8_t*)(t.payload.data() + 4);
+        std::memcpy(&amax, t.data(), 4);
+        const uint8_t* code = (const uint8_t*)(t.data() + 4);
         ks->scratch.resize(n);
         bps_cpu_fp8_decompress(code, n, amax, ks->scratch.data());
         if (first)
@@ -536,10 +649,11 @@ class Server {
         break;
     }
 
-    // push ack
+    // push ack (clear the IPC bit: no 16-byte ext follows the header)
     MsgHeader ack = t.hdr;
     ack.op = kPushReply;
     ack.len = 0;
+    ack.cmd &= ~kCmdIpcPayload;
     t.conn->send(ack, nullptr);
 
     if (ks->async_mode) return;
@@ -560,7 +674,7 @@ class Server {
       auto pending = std::move(ks->pending);
       ks->pending.clear();
       for (auto& p : pending) {
-        if (p.hdr.aux <= ks->version) send_pull_reply(p.conn, p.hdr, ks);
+        if (p.hdr.aux <= ks->version) send_pull_reply(p, ks);
         else ks->pending.push_back(p);
       }
     }

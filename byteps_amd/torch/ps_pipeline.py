@@ -73,16 +73,39 @@ class _KeyInfo:
 
 
 class _Staging:
-    """Pinned host buffers + side stream for one in-flight bucket."""
+    """Host staging + side stream for one in-flight bucket.
 
-    def __init__(self, nbytes_send: int, nbytes_recv: int, device):
+    Colocated server: the buffers are allocated INSIDE a shm region
+    shared with the server (KVClient.ipc_alloc → hipHostRegister'd), so
+    the D2H target IS the server's input and the server's reply lands
+    directly in the H2D source — zero socket copies for gradient bytes.
+    Remote server (or IPC off): private pinned buffers, payload travels
+    inline over TCP."""
+
+    def __init__(self, nbytes_send: int, nbytes_recv: int, device,
+                 kv=None, server: Optional[int] = None):
         self.on_gpu = device.type == "cuda"
-        pin = self.on_gpu
-        self.send = torch.empty(nbytes_send, dtype=torch.uint8,
-                                pin_memory=pin)
-        self.recv = torch.empty(nbytes_recv, dtype=torch.uint8,
-                                pin_memory=pin)
+        self._keep = []
+        self.ipc = False
+        self.send = self._alloc(kv, server, nbytes_send)
+        self.recv = self._alloc(kv, server, nbytes_recv)
         self.stream = torch.cuda.Stream(device) if self.on_gpu else None
+
+    def _alloc(self, kv, server, nbytes: int) -> torch.Tensor:
+        nbytes = max(int(nbytes), 1)
+        if kv is not None and server is not None:
+            try:
+                addr = kv.ipc_alloc(server, nbytes)
+            except Exception:
+                addr = 0
+            if addr:
+                import ctypes
+                buf = (ctypes.c_uint8 * nbytes).from_address(addr)
+                self._keep.append(buf)
+                self.ipc = True
+                return torch.frombuffer(buf, dtype=torch.uint8)
+        return torch.empty(nbytes, dtype=torch.uint8,
+                           pin_memory=self.on_gpu)
 
 
 class Ticket:
@@ -239,7 +262,8 @@ class PSPipeline:
                     cap = max(cap, 64)
                 else:
                     cap = raw
-                s = _Staging(cap, cap, bucket.buffer.device)
+                s = _Staging(cap, cap, bucket.buffer.device,
+                             kv=self.kv, server=ki.server)
                 self._staging[bucket.plan.index] = s
             return s
 
