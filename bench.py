@@ -130,8 +130,8 @@ def main():
     bps.init()
     world = bps.size()
     rank = bps.rank()
-    device = torch.device(args.device, bps.local_rank()) \
-        if args.device == "cuda" else torch.device("cpu")
+    from byteps_amd import common as _C
+    device = _C.device() if args.device == "cuda" else torch.device("cpu")
     on_gpu = device.type == "cuda"
     if on_gpu:
         torch.backends.cudnn.benchmark = True

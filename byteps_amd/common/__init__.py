@@ -91,8 +91,12 @@ def init(lazy: bool = True, backend: Optional[str] = None) -> None:
     cfg.local_rank, cfg.local_size = local_rank, local_size
 
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
-        _state.device = torch.device("cuda", local_rank)
+        # one process per GPU (identity on a full node); modulo keeps
+        # oversubscribed dry-runs alive on smaller boxes (e.g. 2 ranks
+        # sharing the single gpurun GPU with gloo collectives)
+        dev_index = local_rank % max(1, torch.cuda.device_count())
+        torch.cuda.set_device(dev_index)
+        _state.device = torch.device("cuda", dev_index)
     else:
         _state.device = torch.device("cpu")
 
