@@ -74,7 +74,7 @@ class Config:
     force_distributed: bool = False         # force PS path even on 1 node
     enable_async: bool = False              # asynchronous PS training
     enable_mixed_mode: bool = False         # colocated + standalone servers
-    compressor_threads: int = 4
+    compressor_threads: int = 8
     min_compress_bytes: int = 65536
 
     # server knobs ----------------------------------------------------------
@@ -125,7 +125,7 @@ class Config:
         c.enable_mixed_mode = env_bool(
             "BPS_ENABLE_MIXED_MODE", "BYTEPS_ENABLE_MIXED_MODE", default=False)
         c.compressor_threads = env_int(
-            "BPS_COMPRESSOR_THREADS", "BYTEPS_THREADPOOL_SIZE", default=4)
+            "BPS_COMPRESSOR_THREADS", "BYTEPS_THREADPOOL_SIZE", default=8)
         c.min_compress_bytes = env_int(
             "BPS_MIN_COMPRESS_BYTES", "BYTEPS_MIN_COMPRESS_BYTES", default=65536)
 

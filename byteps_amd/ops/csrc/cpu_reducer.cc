@@ -133,6 +133,56 @@ int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
   return 0;
 }
 
+// Fused decode → accumulator (one pass; replaces decompress-to-scratch +
+// copy/sum — the server merge was 3 passes per push, VERDICT round-2
+// item 4: the PS per-step cost lives in these CPU passes).
+int bps_cpu_onebit_accumulate(const uint64_t* bits, float scale_sum,
+                              int64_t n, float* acc, int first) {
+  float scale = scale_sum / (float)n;
+  if (first) {
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i)
+      acc[i] = ((bits[i >> 6] >> (i & 63)) & 1ULL) ? scale : -scale;
+  } else {
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i)
+      acc[i] += ((bits[i >> 6] >> (i & 63)) & 1ULL) ? scale : -scale;
+  }
+  return 0;
+}
+
+// Fused server-side error feedback + sign-pack for the merged reply:
+//   comp = acc + err;  bits = sign(comp);  scale = Σ|comp|/n;
+//   err  = comp − sign(comp)·scale
+// in TWO passes (compensate+pack, then residual) instead of the generic
+// compensate / compress / decompress / subtract four-pass chain.
+int bps_cpu_onebit_reply_ef(const float* acc, float* err, int64_t n,
+                            uint64_t* bits, float* scale_sum, float* comp) {
+  int64_t nwords = (n + 63) >> 6;
+  double l1 = 0.0;
+#pragma omp parallel for reduction(+ : l1)
+  for (int64_t w = 0; w < nwords; ++w) {
+    uint64_t mask = 0;
+    int64_t lim = std::min<int64_t>(64, n - (w << 6));
+    for (int64_t l = 0; l < lim; ++l) {
+      int64_t i = (w << 6) + l;
+      float c = acc[i] + err[i];
+      comp[i] = c;
+      l1 += std::fabs(c);
+      if (c >= 0.0f) mask |= (1ULL << l);
+    }
+    bits[w] = mask;
+  }
+  float scale = (float)(l1 / (double)n);
+  *scale_sum = (float)l1;
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) {
+    float c = comp[i];
+    err[i] = c - (c >= 0.0f ? scale : -scale);
+  }
+  return 0;
+}
+
 int bps_cpu_randomk_indices(int64_t n, int64_t k, uint64_t seed,
                             int32_t* idx) {
   // regenerate the worker's draws from the seed (counter-mode)

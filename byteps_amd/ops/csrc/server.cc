@@ -63,6 +63,10 @@ int bps_cpu_onebit_compress(const float* x, int64_t n, uint64_t* bits,
                             float* scale_sum);
 int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
                               float* out);
+int bps_cpu_onebit_accumulate(const uint64_t* bits, float scale_sum,
+                              int64_t n, float* acc, int first);
+int bps_cpu_onebit_reply_ef(const float* acc, float* err, int64_t n,
+                            uint64_t* bits, float* scale_sum, float* comp);
 int bps_cpu_sparse_accumulate(const int32_t* idx, const float* val, int64_t k,
                               float* acc);
 int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
@@ -600,13 +604,10 @@ class Server {
         int64_t nwords = (n + 63) >> 6;
         float scale_sum;
         std::memcpy(&scale_sum, t.data() + nwords * 8, 4);
-        ks->scratch.resize(n);
-        bps_cpu_onebit_decompress((const uint64_t*)t.data(),
-                                  scale_sum, n, ks->scratch.data());
-        if (first)
-          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
-        else
-          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        // fused decode→accumulate: one pass instead of
+        // decompress-to-scratch + copy/sum
+        bps_cpu_onebit_accumulate((const uint64_t*)t.data(), scale_sum, n,
+                                  acc, first ? 1 : 0);
         break;
       }
       case kTopk:
@@ -683,6 +684,21 @@ class Server {
   void compress_reply(KeyState* ks) {
     const int64_t n = (int64_t)ks->nelem;
     const float* acc = ks->store.data();
+    if (ks->server_ef && ks->codec == kOnebit) {
+      // fused EF + sign-pack: two passes instead of the generic
+      // compensate/compress/decompress/subtract chain
+      if (ks->ef_err.empty()) ks->ef_err.assign(n, 0.0f);
+      ks->ef_comp.resize(n);
+      int64_t nwords = (n + 63) >> 6;
+      ks->reply.resize(nwords * 8 + 8);
+      float scale_sum = 0.0f;
+      bps_cpu_onebit_reply_ef(acc, ks->ef_err.data(), n,
+                              (uint64_t*)ks->reply.data(), &scale_sum,
+                              ks->ef_comp.data());
+      std::memcpy(ks->reply.data() + nwords * 8, &scale_sum, 4);
+      std::memset(ks->reply.data() + nwords * 8 + 4, 0, 4);
+      return;
+    }
     // server-side error feedback (reference server compressor mirror,
     // server/server.cc:228-257 + vanilla EF): compensate the merge with
     // the previous reply's residual before compressing, then store the
