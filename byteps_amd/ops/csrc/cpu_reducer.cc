@@ -151,13 +151,17 @@ int bps_cpu_onebit_accumulate(const uint64_t* bits, float scale_sum,
   return 0;
 }
 
-// Fused server-side error feedback + sign-pack for the merged reply:
-//   comp = acc + err;  bits = sign(comp);  scale = Σ|comp|/n;
-//   err  = comp − sign(comp)·scale
-// in TWO passes (compensate+pack, then residual) instead of the generic
-// compensate / compress / decompress / subtract four-pass chain.
-int bps_cpu_onebit_reply_ef(const float* acc, float* err, int64_t n,
-                            uint64_t* bits, float* scale_sum, float* comp) {
+// Fused server-side error feedback + sign-pack for the merged reply,
+// split so only pass 1 sits on the pull critical path:
+//   pass 1 (reply_pack): comp = acc + err;  bits = sign(comp);
+//                        scale_sum = Σ|comp|       → reply is ready
+//   pass 2 (err_update): err = comp − sign(comp)·scale   → runs AFTER
+//                        the queued pulls flush; only needed next round
+// (replaces the generic compensate/compress/decompress/subtract
+// four-pass chain).
+int bps_cpu_onebit_reply_pack(const float* acc, const float* err, int64_t n,
+                              uint64_t* bits, float* scale_sum,
+                              float* comp) {
   int64_t nwords = (n + 63) >> 6;
   double l1 = 0.0;
 #pragma omp parallel for reduction(+ : l1)
@@ -173,8 +177,13 @@ int bps_cpu_onebit_reply_ef(const float* acc, float* err, int64_t n,
     }
     bits[w] = mask;
   }
-  float scale = (float)(l1 / (double)n);
   *scale_sum = (float)l1;
+  return 0;
+}
+
+int bps_cpu_onebit_err_update(const float* comp, int64_t n, float scale_sum,
+                              float* err) {
+  float scale = scale_sum / (float)n;
 #pragma omp parallel for
   for (int64_t i = 0; i < n; ++i) {
     float c = comp[i];

@@ -65,8 +65,10 @@ int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
                               float* out);
 int bps_cpu_onebit_accumulate(const uint64_t* bits, float scale_sum,
                               int64_t n, float* acc, int first);
-int bps_cpu_onebit_reply_ef(const float* acc, float* err, int64_t n,
-                            uint64_t* bits, float* scale_sum, float* comp);
+int bps_cpu_onebit_reply_pack(const float* acc, const float* err, int64_t n,
+                              uint64_t* bits, float* scale_sum, float* comp);
+int bps_cpu_onebit_err_update(const float* comp, int64_t n, float scale_sum,
+                              float* err);
 int bps_cpu_sparse_accumulate(const int32_t* idx, const float* val, int64_t k,
                               float* acc);
 int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
@@ -678,6 +680,17 @@ class Server {
         if (p.hdr.aux <= ks->version) send_pull_reply(p, ks);
         else ks->pending.push_back(p);
       }
+      if (ks->server_ef && ks->codec == kOnebit) {
+        // deferred EF residual: only needed by NEXT round's pack, so it
+        // runs after the queued pulls were answered (off the pull
+        // critical path)
+        int64_t nn = (int64_t)ks->nelem;
+        int64_t nwords = (nn + 63) >> 6;
+        float sc;
+        std::memcpy(&sc, ks->reply.data() + nwords * 8, 4);
+        bps_cpu_onebit_err_update(ks->ef_comp.data(), nn, sc,
+                                  ks->ef_err.data());
+      }
     }
   }
 
@@ -692,9 +705,9 @@ class Server {
       int64_t nwords = (n + 63) >> 6;
       ks->reply.resize(nwords * 8 + 8);
       float scale_sum = 0.0f;
-      bps_cpu_onebit_reply_ef(acc, ks->ef_err.data(), n,
-                              (uint64_t*)ks->reply.data(), &scale_sum,
-                              ks->ef_comp.data());
+      bps_cpu_onebit_reply_pack(acc, ks->ef_err.data(), n,
+                                (uint64_t*)ks->reply.data(), &scale_sum,
+                                ks->ef_comp.data());
       std::memcpy(ks->reply.data() + nwords * 8, &scale_sum, 4);
       std::memset(ks->reply.data() + nwords * 8 + 4, 0, 4);
       return;

@@ -6,11 +6,14 @@
 Run on a GPU box:  python scripts/micro_epilogue.py
 """
 
+import os
+import sys
 import time
 
 import torch
 
-from byteps_amd import ops as K
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from byteps_amd import ops as K  # noqa: E402
 
 
 def bench(fn, iters=200):
