@@ -139,7 +139,16 @@ class RandomkCompressor(BaseCompressor):
 
 class DitheringCompressor(BaseCompressor):
     """Stochastic quantization, linear or natural (power-of-2) partitions
-    (reference impl/dithering.cc)."""
+    (reference impl/dithering.cc).
+
+    Wire: ``[norm f32][flag u8][body]`` — flag 0 is dense int8 codes,
+    flag 1 is the chunked Elias-delta sparse stream (reference
+    utils.h:115-250 BitWriter + delta-coded positions).  Quantization
+    runs on the GPU; the bit-level (de)coding is a host-side transform
+    between staging and the wire (``encode_wire``/``decode_wire``),
+    exactly where the reference ran its whole codec."""
+
+    host_wire = True
 
     def __init__(self, s: int = 64, natural: bool = False, seed: int = 1):
         self.s = int(s)
@@ -148,17 +157,51 @@ class DitheringCompressor(BaseCompressor):
         self.codec = DITHER_NATURAL if natural else DITHER_LINEAR
         self.seed = seed
         self.round = 0
+        from ..common.config import env_bool
+        self.sparse_wire = env_bool("BPS_DITHER_SPARSE", default=True)
 
     def compress(self, x: torch.Tensor) -> Compressed:
         seed = (self.seed * 0xD6E8FEB86659FD93 + self.round) & (2**64 - 1)
         self.round += 1
         code, norm_t = K.dithering_compress(x.reshape(-1), self.s, seed,
                                             self.natural)
-        return Compressed([norm_t.view(torch.uint8), code.view(torch.uint8)])
+        flag = torch.zeros(1, dtype=torch.uint8, device=x.device)
+        return Compressed([norm_t.view(torch.uint8), flag,
+                           code.view(torch.uint8)])
+
+    def encode_wire(self, host_payload: torch.Tensor, n: int,
+                    out_buf) -> int:
+        """Dense host payload → Elias wire in ``out_buf``.  Returns the
+        wire length, or -1 when the dense form is smaller (caller pushes
+        the dense payload unchanged)."""
+        if not self.sparse_wire or out_buf is None:
+            return -1
+        from ..ops import core
+        wlen = core().cpu_dither_encode(
+            host_payload.data_ptr() + 5, n, out_buf.data_ptr() + 5,
+            min(out_buf.numel(), host_payload.numel()) - 5)
+        if wlen < 0:
+            return -1
+        out_buf[:4] = host_payload[:4]
+        out_buf[4] = 1
+        return 5 + int(wlen)
+
+    def decode_wire(self, wire: torch.Tensor, n: int) -> torch.Tensor:
+        """Sparse host wire → dense host payload (no-op for flag 0)."""
+        if int(wire[4]) != 1:
+            return wire
+        from ..ops import core
+        dense = torch.empty(5 + n, dtype=torch.uint8)
+        dense[:4] = wire[:4]
+        dense[4] = 0
+        body = wire[5:].contiguous()
+        core().cpu_dither_decode(body.data_ptr(), body.numel(), n,
+                                 dense.data_ptr() + 5)
+        return dense
 
     def decompress(self, payload, n, aux=0, out=None):
         norm_t = payload[:4].view(torch.float32).contiguous()
-        code = payload[4:4 + n].view(torch.int8).contiguous()
+        code = payload[5:5 + n].view(torch.int8).contiguous()
         return K.dithering_decompress(code, norm_t, self.s, self.natural, out)
 
 

@@ -177,34 +177,44 @@ __global__ void dithering_linear_decompress_kernel(
     out[i] = (float)code[i] / (float)s * nrm;
 }
 
-// natural: levels are powers of two.  r ∈ (0,1] lies in [2^e, 2^(e+1));
-// round stochastically to an endpoint, store signed biased exponent:
-// code = sign * (e + 128 + 1) clamped; 0 ⇒ value 0.  |e| ≤ 126.
+// natural: s power-of-two levels {2^(1-s), …, 2^-1, 2^0}·norm (reference
+// impl/dithering.cc natural partitions).  r ∈ [2^e, 2^(e+1)) rounds
+// stochastically to an endpoint; r below the lowest level rounds to 0 or
+// 2^(1-s) with p = r/2^(1-s) — THIS is where the sparsity that makes the
+// Elias wire pay comes from.  code = sign·(e + s) ∈ [-s, s]; 0 ⇒ 0.
 __global__ void dithering_natural_compress_kernel(
-    const float* __restrict__ x, int64_t n, uint64_t seed,
+    const float* __restrict__ x, int64_t n, int s, uint64_t seed,
     const float* __restrict__ norm, int8_t* __restrict__ code) {
   const float nrm = norm[0];
+  const float lowest = ldexpf(1.0f, 1 - s);
   int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * BLOCK;
   for (int64_t i = i0; i < n; i += stride) {
     float v = x[i];
     float r = (nrm > 0.0f) ? fabsf(v) / nrm : 0.0f;
     if (r <= 0.0f) { code[i] = 0; continue; }
-    int e;
-    float m = frexpf(r, &e);         // r = m * 2^e, m ∈ [0.5, 1)
-    // interval endpoints: lo = 2^(e-1), hi = 2^e ; p(up) = (r-lo)/(lo)
-    float p_up = m * 2.0f - 1.0f;    // (r - lo)/lo
-    int ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
-    if (ebits < -120) { code[i] = 0; continue; }
-    if (ebits > 0) ebits = 0;        // r ≤ 1 ⇒ level ≤ 2^0
-    int biased = ebits + 121;        // ∈ [1, 121]
+    int ebits;
+    if (r < lowest) {
+      float p_up = r / lowest;
+      if (uniform_at(seed, (uint64_t)i) >= p_up) { code[i] = 0; continue; }
+      ebits = 1 - s;
+    } else {
+      int e;
+      float m = frexpf(r, &e);       // r = m * 2^e, m ∈ [0.5, 1)
+      // interval endpoints: lo = 2^(e-1), hi = 2^e ; p(up) = (r-lo)/lo
+      float p_up = m * 2.0f - 1.0f;
+      ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
+      if (ebits > 0) ebits = 0;      // r ≤ 1 ⇒ level ≤ 2^0
+      if (ebits < 1 - s) ebits = 1 - s;
+    }
+    int biased = ebits + s;          // ∈ [1, s]
     code[i] = (int8_t)(v < 0.0f ? -biased : biased);
   }
 }
 
 __global__ void dithering_natural_decompress_kernel(
-    const int8_t* __restrict__ code, int64_t n, const float* __restrict__ norm,
-    float* __restrict__ out) {
+    const int8_t* __restrict__ code, int64_t n, int s,
+    const float* __restrict__ norm, float* __restrict__ out) {
   const float nrm = norm[0];
   int64_t i0 = (int64_t)blockIdx.x * BLOCK + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * BLOCK;
@@ -212,7 +222,7 @@ __global__ void dithering_natural_decompress_kernel(
     int c = code[i];
     if (c == 0) { out[i] = 0.0f; continue; }
     int mag = c < 0 ? -c : c;
-    float v = ldexpf(1.0f, mag - 121) * nrm;
+    float v = ldexpf(1.0f, mag - s) * nrm;
     out[i] = c < 0 ? -v : v;
   }
 }
@@ -314,7 +324,7 @@ int bps_dithering_compress(const void* x, int64_t n, int s, uint64_t seed,
                            void* stream) {
   if (natural)
     hipLaunchKernelGGL(dithering_natural_compress_kernel, dim3(grid_for(n)),
-                       dim3(BLOCK), 0, STREAM, (const float*)x, n, seed,
+                       dim3(BLOCK), 0, STREAM, (const float*)x, n, s, seed,
                        (const float*)norm, (int8_t*)code);
   else
     hipLaunchKernelGGL(dithering_linear_compress_kernel, dim3(grid_for(n)),
@@ -327,7 +337,7 @@ int bps_dithering_decompress(const void* code, int64_t n, int s, int natural,
                              const void* norm, void* out, void* stream) {
   if (natural)
     hipLaunchKernelGGL(dithering_natural_decompress_kernel, dim3(grid_for(n)),
-                       dim3(BLOCK), 0, STREAM, (const int8_t*)code, n,
+                       dim3(BLOCK), 0, STREAM, (const int8_t*)code, n, s,
                        (const float*)norm, (float*)out);
   else
     hipLaunchKernelGGL(dithering_linear_decompress_kernel, dim3(grid_for(n)),

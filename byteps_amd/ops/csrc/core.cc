@@ -97,6 +97,10 @@ int bps_cpu_copy(void* dst, const void* src, int64_t nbytes);
 int bps_cpu_scale(void* x, int64_t n, float alpha, int dtype);
 int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
                         float* val);
+int bps_cpu_dither_encode(const int8_t* code, int64_t n, uint8_t* out,
+                          int64_t out_cap, int64_t* out_len);
+int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
+                          int8_t* code);
 int bps_cpu_onebit_compress(const float* x, int64_t n, uint64_t* bits,
                             float* scale_sum);
 int bps_cpu_onebit_decompress(const uint64_t* bits, float scale_sum, int64_t n,
@@ -330,6 +334,19 @@ PYBIND11_MODULE(_core, m) {
   m.def("cpu_scale", [](uintptr_t x, int64_t n, float a, int dt) {
     check(bps_cpu_scale(P(x), n, a, dt), "bps_cpu_scale");
   }, py::call_guard<py::gil_scoped_release>());
+  m.def("cpu_dither_encode",
+        [](uintptr_t code, int64_t n, uintptr_t out, int64_t cap) {
+          int64_t wlen = 0;
+          int rc = bps_cpu_dither_encode((const int8_t*)CP(code), n,
+                                         (uint8_t*)P(out), cap, &wlen);
+          return rc == 0 ? wlen : (int64_t)-1;
+        }, py::call_guard<py::gil_scoped_release>());
+  m.def("cpu_dither_decode",
+        [](uintptr_t in, int64_t in_len, int64_t n, uintptr_t code) {
+          check(bps_cpu_dither_decode((const uint8_t*)CP(in), in_len, n,
+                                      (int8_t*)P(code)),
+                "bps_cpu_dither_decode");
+        }, py::call_guard<py::gil_scoped_release>());
   m.def("cpu_topk_select",
         [](uintptr_t x, int64_t n, int64_t k, uintptr_t idx, uintptr_t val) {
           check(bps_cpu_topk_select((const float*)CP(x), n, k,

@@ -10,6 +10,7 @@
 #include <cmath>
 #include <cstdint>
 #include <cstring>
+#include <atomic>
 #include <vector>
 
 #ifdef _OPENMP
@@ -228,20 +229,30 @@ int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
       if (level > s) level = s;
       code[i] = (int8_t)(v < 0.0f ? -level : level);
     } else {
+      // s power-of-two levels; below 2^(1-s) stochastically → 0 (the
+      // sparsity source) — bit-identical twin of the gfx950 kernel
       if (r <= 0.0f) {
         code[i] = 0;
         continue;
       }
-      int e;
-      float m = std::frexp(r, &e);
-      float p_up = m * 2.0f - 1.0f;
-      int ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
-      if (ebits < -120) {
-        code[i] = 0;
-        continue;
+      float lowest = std::ldexp(1.0f, 1 - s);
+      int ebits;
+      if (r < lowest) {
+        float p_up = r / lowest;
+        if (uniform_at(seed, (uint64_t)i) >= p_up) {
+          code[i] = 0;
+          continue;
+        }
+        ebits = 1 - s;
+      } else {
+        int e;
+        float m = std::frexp(r, &e);
+        float p_up = m * 2.0f - 1.0f;
+        ebits = e - 1 + ((uniform_at(seed, (uint64_t)i) < p_up) ? 1 : 0);
+        if (ebits > 0) ebits = 0;
+        if (ebits < 1 - s) ebits = 1 - s;
       }
-      if (ebits > 0) ebits = 0;
-      int biased = ebits + 121;
+      int biased = ebits + s;
       code[i] = (int8_t)(v < 0.0f ? -biased : biased);
     }
   }
@@ -261,7 +272,7 @@ int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
         continue;
       }
       int mag = c < 0 ? -c : c;
-      float v = std::ldexp(1.0f, mag - 121) * norm;
+      float v = std::ldexp(1.0f, mag - s) * norm;
       out[i] = c < 0 ? -v : v;
     }
   }
@@ -283,6 +294,196 @@ int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
 #pragma omp parallel for
   for (int64_t i = 0; i < n; ++i)
     out[i] = fp8_e4m3_decode(code[i]) * inv;
+  return 0;
+}
+
+// ---------------------------------------------------------------------------
+// Elias-delta sparse wire for dithering codes (reference
+// common/compressor/utils.h:115-250 BitWriter/Elias-delta +
+// impl/dithering.cc:51-121 sign/position coding — re-designed as a
+// CHUNKED stream so encode and decode parallelize):
+//
+//   [u32 chunk_elems][u32 nchunks]
+//   nchunks × [u32 nbytes]                  (chunk byte sizes)
+//   nchunks × chunk streams, byte aligned:
+//       per nonzero, MSB-first bits: elias_delta(gap+1) ·
+//       sign bit (1 ⇒ negative) · elias_delta(|code|) ; then
+//       elias_delta(chunk_remainder+1) terminates the chunk.
+//
+// elias_delta(x ≥ 1): x = 2^(N-1)+rest; N = 2^(L-1)+… encoded as
+// L-1 zeros, bits of N, then N-1 low bits of x.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int64_t kChunkElems = 1 << 16;
+
+struct BitWriter {
+  uint8_t* p;
+  uint64_t bitpos = 0;
+  explicit BitWriter(uint8_t* out) : p(out) {}
+  inline void put_bit(int b) {
+    if (b) p[bitpos >> 3] |= (uint8_t)(0x80u >> (bitpos & 7));
+    bitpos++;
+  }
+  inline void put_bits(uint32_t v, int nbits) {  // MSB first
+    for (int i = nbits - 1; i >= 0; --i) put_bit((v >> i) & 1);
+  }
+  inline void elias_delta(uint32_t x) {  // x >= 1
+    int nb = 31;
+    while (nb > 0 && !((x >> nb) & 1)) --nb;       // nb = floor(log2 x)
+    uint32_t N = (uint32_t)nb + 1;
+    int lb = 31;
+    while (lb > 0 && !((N >> lb) & 1)) --lb;       // lb = floor(log2 N)
+    for (int i = 0; i < lb; ++i) put_bit(0);
+    put_bits(N, lb + 1);
+    if (nb > 0) put_bits(x & ((1u << nb) - 1), nb);
+  }
+  inline int64_t bytes() const { return (int64_t)((bitpos + 7) >> 3); }
+};
+
+struct BitReader {
+  const uint8_t* p;
+  uint64_t bitpos = 0;
+  explicit BitReader(const uint8_t* in) : p(in) {}
+  inline int get_bit() {
+    int b = (p[bitpos >> 3] >> (7 - (bitpos & 7))) & 1;
+    bitpos++;
+    return b;
+  }
+  inline uint32_t get_bits(int nbits) {
+    uint32_t v = 0;
+    for (int i = 0; i < nbits; ++i) v = (v << 1) | get_bit();
+    return v;
+  }
+  inline uint32_t elias_delta() {
+    int lb = 0;
+    while (!get_bit()) ++lb;
+    uint32_t N = (1u << lb) | get_bits(lb);
+    int nb = (int)N - 1;
+    uint32_t x = 1u << nb;
+    if (nb > 0) x |= get_bits(nb);
+    return x;
+  }
+};
+
+// encode one chunk of codes; returns bytes written
+int64_t encode_chunk(const int8_t* code, int64_t n, uint8_t* out) {
+  BitWriter w(out);
+  int64_t prev = -1;
+  for (int64_t i = 0; i < n; ++i) {
+    if (code[i] == 0) continue;
+    w.elias_delta((uint32_t)(i - prev));  // gap+1 ≥ 1
+    w.put_bit(code[i] < 0);
+    w.elias_delta((uint32_t)(code[i] < 0 ? -code[i] : code[i]));
+    prev = i;
+  }
+  w.elias_delta((uint32_t)(n - prev));    // terminator: remainder+1
+  return w.bytes();
+}
+
+}  // namespace
+
+// Encode dithering codes → sparse wire.  Returns total bytes, or -1 if
+// out_cap is too small (caller falls back to the dense int8 wire).
+int bps_cpu_dither_encode(const int8_t* code, int64_t n, uint8_t* out,
+                          int64_t out_cap, int64_t* out_len) {
+  int64_t nchunks = (n + kChunkElems - 1) / kChunkElems;
+  if (nchunks == 0) nchunks = 1;
+  int64_t header = 8 + 4 * nchunks;
+  if (out_cap < header) return -1;
+  // worst case ~14 bits/elem + terminator
+  std::vector<std::vector<uint8_t>> bufs(nchunks);
+  std::vector<int64_t> sizes(nchunks, 0);
+  bool overflow = false;
+#pragma omp parallel for schedule(dynamic)
+  for (int64_t c = 0; c < nchunks; ++c) {
+    int64_t lo = c * kChunkElems;
+    int64_t len = std::min<int64_t>(kChunkElems, n - lo);
+    bufs[c].assign((size_t)(len * 2 + 16), 0);
+    sizes[c] = encode_chunk(code + lo, len, bufs[c].data());
+  }
+  int64_t total = header;
+  for (int64_t c = 0; c < nchunks; ++c) total += sizes[c];
+  if (overflow || total > out_cap) return -1;
+  uint32_t ce = (uint32_t)kChunkElems, nc = (uint32_t)nchunks;
+  std::memcpy(out, &ce, 4);
+  std::memcpy(out + 4, &nc, 4);
+  for (int64_t c = 0; c < nchunks; ++c) {
+    uint32_t sz = (uint32_t)sizes[c];
+    std::memcpy(out + 8 + 4 * c, &sz, 4);
+  }
+  int64_t off = header;
+  for (int64_t c = 0; c < nchunks; ++c) {
+    std::memcpy(out + off, bufs[c].data(), sizes[c]);
+    off += sizes[c];
+  }
+  *out_len = total;
+  return 0;
+}
+
+// Decode sparse wire → dense codes (zeros included).  Returns 0, or -1
+// on malformed input.
+int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
+                          int8_t* code) {
+  if (in_len < 8) return -1;
+  uint32_t ce, nc;
+  std::memcpy(&ce, in, 4);
+  std::memcpy(&nc, in + 4, 4);
+  if (ce == 0 || nc == 0 || (int64_t)8 + 4 * nc > in_len) return -1;
+  std::vector<int64_t> offs(nc + 1, 0);
+  offs[0] = 8 + 4 * (int64_t)nc;
+  for (uint32_t c = 0; c < nc; ++c) {
+    uint32_t sz;
+    std::memcpy(&sz, in + 8 + 4 * c, 4);
+    offs[c + 1] = offs[c] + sz;
+  }
+  if (offs[nc] > in_len) return -1;
+  std::memset(code, 0, (size_t)n);
+  std::atomic<int> bad{0};
+#pragma omp parallel for schedule(dynamic)
+  for (int64_t c = 0; c < (int64_t)nc; ++c) {
+    int64_t lo = (int64_t)c * ce;
+    int64_t len = std::min<int64_t>(ce, n - lo);
+    if (len <= 0) continue;
+    BitReader r(in + offs[c]);
+    uint64_t maxbits = (uint64_t)(offs[c + 1] - offs[c]) * 8;
+    int64_t i = -1;
+    for (;;) {
+      if (r.bitpos >= maxbits) { bad = 1; break; }
+      uint32_t gap = r.elias_delta();
+      i += gap;
+      if (i >= len) break;           // terminator
+      int neg = r.get_bit();
+      uint32_t mag = r.elias_delta();
+      if (mag > 127) { bad = 1; break; }
+      code[lo + i] = (int8_t)(neg ? -(int)mag : (int)mag);
+    }
+  }
+  return bad ? -1 : 0;
+}
+
+// Fused dithering decode → accumulator (server push path, one pass over
+// dense codes)
+int bps_cpu_dithering_accumulate(const int8_t* code, int64_t n, int s,
+                                 int natural, float norm, float* acc,
+                                 int first) {
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) {
+    int c = code[i];
+    float v;
+    if (!natural) {
+      v = (float)c / (float)s * norm;
+    } else if (c == 0) {
+      v = 0.0f;
+    } else {
+      int mag = c < 0 ? -c : c;
+      v = std::ldexp(1.0f, mag - s) * norm;
+      if (c < 0) v = -v;
+    }
+    if (first) acc[i] = v;
+    else acc[i] += v;
+  }
   return 0;
 }
 

@@ -75,6 +75,13 @@ int bps_cpu_dithering_compress(const float* x, int64_t n, int s, uint64_t seed,
                                int natural, float norm, int8_t* code);
 int bps_cpu_dithering_decompress(const int8_t* code, int64_t n, int s,
                                  int natural, float norm, float* out);
+int bps_cpu_dithering_accumulate(const int8_t* code, int64_t n, int s,
+                                 int natural, float norm, float* acc,
+                                 int first);
+int bps_cpu_dither_encode(const int8_t* code, int64_t n, uint8_t* out,
+                          int64_t out_cap, int64_t* out_len);
+int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
+                          int8_t* code);
 int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
                         float* val);
 int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
@@ -205,6 +212,7 @@ struct KeyState {
   std::vector<PendingPull> pending;
   uint64_t push_total = 0;       // scheduling signal
   std::vector<float> scratch;    // decompress workspace
+  std::vector<int8_t> code_scratch;  // dithering dense-code workspace
   bool server_ef = false;        // error-feedback on the merged reply
   std::vector<float> ef_err;     // residual of the previous reply
   std::vector<float> ef_comp;    // compensated merge workspace
@@ -623,17 +631,24 @@ class Server {
       }
       case kDitherLinear:
       case kDitherNatural: {
+        // wire: [norm f32][flag u8][dense int8 | Elias-delta stream]
         float norm;
         std::memcpy(&norm, t.data(), 4);
-        const int8_t* code = (const int8_t*)(t.data() + 4);
-        ks->scratch.resize(n);
-        bps_cpu_dithering_decompress(code, n, (int)ks->levels,
-                                     codec == kDitherNatural, norm,
-                                     ks->scratch.data());
-        if (first)
-          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
-        else
-          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        uint8_t flag = (uint8_t)t.data()[4];
+        const int8_t* code;
+        if (flag == 1) {
+          ks->code_scratch.resize(n);
+          if (bps_cpu_dither_decode((const uint8_t*)t.data() + 5,
+                                    (int64_t)t.hdr.len - 5, n,
+                                    ks->code_scratch.data()) != 0)
+            break;  // malformed — drop (worker sees stalled round)
+          code = ks->code_scratch.data();
+        } else {
+          code = (const int8_t*)(t.data() + 5);
+        }
+        bps_cpu_dithering_accumulate(code, n, (int)ks->levels,
+                                     codec == kDitherNatural, norm, acc,
+                                     first ? 1 : 0);
         break;
       }
       case kFp8: {
@@ -763,10 +778,21 @@ class Server {
         bool natural = ks->codec == kDitherNatural;
         float norm = bps_cpu_norm(acc, n, natural ? 1 : 2);
         uint64_t seed = splitmix64(ks->version * 0xD6E8FEB86659FD93ULL + 5);
-        ks->reply.resize(4 + n);
-        std::memcpy(ks->reply.data(), &norm, 4);
+        ks->code_scratch.resize(n);
         bps_cpu_dithering_compress(acc, n, (int)ks->levels, seed, natural,
-                                   norm, (int8_t*)(ks->reply.data() + 4));
+                                   norm, ks->code_scratch.data());
+        ks->reply.resize(5 + n);
+        std::memcpy(ks->reply.data(), &norm, 4);
+        int64_t wlen = 0;
+        if (bps_cpu_dither_encode(ks->code_scratch.data(), n,
+                                  (uint8_t*)ks->reply.data() + 5, n,
+                                  &wlen) == 0) {
+          ks->reply[4] = 1;          // Elias sparse wire
+          ks->reply.resize(5 + wlen);
+        } else {
+          ks->reply[4] = 0;          // dense fallback (stream ≥ dense)
+          std::memcpy(ks->reply.data() + 5, ks->code_scratch.data(), n);
+        }
         break;
       }
       case kFp8: {
@@ -806,8 +832,9 @@ class Server {
         case kDitherNatural: {
           float norm;
           std::memcpy(&norm, ks->reply.data(), 4);
+          // dense codes of the reply are still in code_scratch
           bps_cpu_dithering_decompress(
-              (const int8_t*)(ks->reply.data() + 4), n, (int)ks->levels,
+              ks->code_scratch.data(), n, (int)ks->levels,
               ks->codec == kDitherNatural, norm, dec);
           break;
         }

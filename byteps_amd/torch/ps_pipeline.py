@@ -83,12 +83,16 @@ class _Staging:
     inline over TCP."""
 
     def __init__(self, nbytes_send: int, nbytes_recv: int, device,
-                 kv=None, server: Optional[int] = None):
+                 kv=None, server: Optional[int] = None,
+                 wire_bytes: int = 0):
         self.on_gpu = device.type == "cuda"
         self._keep = []
         self.ipc = False
         self.send = self._alloc(kv, server, nbytes_send)
         self.recv = self._alloc(kv, server, nbytes_recv)
+        # host-side wire transform output (Elias-coded dithering)
+        self.wire = self._alloc(kv, server, wire_bytes) if wire_bytes \
+            else None
         self.stream = torch.cuda.Stream(device) if self.on_gpu else None
 
     def _alloc(self, kv, server, nbytes: int) -> torch.Tensor:
@@ -251,19 +255,25 @@ class PSPipeline:
             s = self._staging.get(bucket.plan.index)
             if s is None:
                 raw = ki.nelem * 4
+                wire_bytes = 0
                 if ki.compressor is not None:
                     codec = ki.compressor.codec
                     if codec in (2, 3):        # topk / randomk: 8 B per pair
                         cap = 8 * min(max(ki.compressor.levels, 1), ki.nelem)
                     elif codec == 1:           # onebit: bits + scale
                         cap = ((ki.nelem + 63) // 64) * 8 + 8
-                    else:                      # dithering: norm + int8 codes
+                    elif codec in (4, 5):      # dithering: norm+flag+codes
+                        cap = ki.nelem + 5
+                        if getattr(ki.compressor, "host_wire", False):
+                            wire_bytes = cap
+                    else:                      # fp8: amax + codes
                         cap = ki.nelem + 4
                     cap = max(cap, 64)
                 else:
                     cap = raw
                 s = _Staging(cap, cap, bucket.buffer.device,
-                             kv=self.kv, server=ki.server)
+                             kv=self.kv, server=ki.server,
+                             wire_bytes=wire_bytes)
                 self._staging[bucket.plan.index] = s
             return s
 
@@ -375,6 +385,15 @@ class PSPipeline:
                 st.stream.synchronize()
 
         _tr("compress+d2h", False)
+        # host-side wire transform (Elias-coded dithering): encode the
+        # dense staging into the wire buffer; fall back to dense when the
+        # stream would be larger
+        push_buf = st.send
+        if comp is not None and getattr(comp, "host_wire", False):
+            wlen = comp.encode_wire(st.send[:nbytes], ki.nelem, st.wire)
+            if wlen > 0:
+                push_buf = st.wire
+                nbytes = wlen
         codec = comp.codec if comp is not None else 0
         cmd = _make_cmd(codec, 0, self.cfg.enable_async)
         # full duplex: push and pull are both in flight — the server
@@ -384,7 +403,7 @@ class PSPipeline:
         # docs/faq.md:23-25)
         _tr("push", True)
         t_push = self.kv.submit(ki.server, _OP_PUSH, ki.key,
-                                st.send.data_ptr(), nbytes, 0, 0, cmd,
+                                push_buf.data_ptr(), nbytes, 0, 0, cmd,
                                 push_aux)
         t_pull = self.kv.submit(ki.server, _OP_PULL, ki.key, 0, 0,
                                 st.recv.data_ptr(), st.recv.numel(), cmd,
@@ -409,6 +428,9 @@ class PSPipeline:
         with stream_ctx:
             if comp is not None:
                 wire = st.recv[:reply_len]
+                if getattr(comp, "host_wire", False):
+                    # Elias reply → dense host payload before H2D
+                    wire = comp.decode_wire(wire, ki.nelem)
                 if on_gpu:
                     wire = wire.to(shard.device, non_blocking=True)
                 # sparse codecs: the server's reply k = min(levels, nelem);
