@@ -151,10 +151,59 @@ class GradEngine:
         for name in self.param_names:
             C._state.registry.declare("byteps.Gradient." + name)
 
-        dt = grad_dtype or self.params[0].dtype
-        dev = self.params[0].device
+        self._grad_dt = grad_dtype or self.params[0].dtype
+        self._device = self.params[0].device
         part_bytes = partition_bytes or self.cfg.partition_bytes
-        part_elems = max(part_bytes // torch.tensor([], dtype=dt).element_size(), 4096)
+        self._part_elems = max(
+            part_bytes // torch.tensor([], dtype=self._grad_dt).element_size(),
+            4096)
+        self._generation = 0
+        self._build_buckets()
+        dt = self._grad_dt
+
+        # optional reduced-precision wire for the RCCL collectives
+        # (``comm_dtype`` argument — set by Compression.fp16/bf16 — or
+        # BPS_COMM_DTYPE=bf16): cast bucket → persistent 16-bit scratch,
+        # all-reduce the scratch, cast back at synchronize.  Halves xGMI
+        # bytes; gradients still ACCUMULATE at full precision — only the
+        # wire is narrowed, matching the reference's Compression semantics
+        # (torch/compression.py:34-76).
+        if comm_dtype is None:
+            from ..common.config import env_str
+            wire = env_str("BPS_COMM_DTYPE", default="").lower()
+            comm_dtype = torch.bfloat16 if wire in ("bf16", "bfloat16") \
+                else torch.float16 if wire in ("fp16", "half") else None
+        self.comm_dtype = comm_dtype
+        if self.comm_dtype is not None and dt == self.comm_dtype:
+            self.comm_dtype = None
+        self._wire_scratch: Dict[int, torch.Tensor] = {}
+        if self.comm_dtype is not None:
+            # eager: the fused cast+scale descriptor needs stable pointers
+            for b in self.buckets:
+                self._wire_scratch[b.plan.index] = torch.empty_like(
+                    b.buffer, dtype=self.comm_dtype)
+        self._fused_desc = None     # (desc dev tensor, total_vec, vec)
+
+        self._ps = None
+        if C._state.ps_enabled:
+            from . import ps_pipeline
+            self._ps = ps_pipeline.get_pipeline(self)
+
+        self._attach_hooks()
+        log.debug("GradEngine: %d params → %d buckets (%.1f MiB each max)",
+                  len(self.params), len(self.buckets),
+                  self._part_elems * self.buckets[0].buffer.element_size()
+                  / 2**20)
+
+    # -- bucket plan (built at init; rebuilt on elastic re-bucketing) ------
+
+    def _build_buckets(self) -> None:
+        """Pack params into flat aligned buckets and alias ``p.grad``.
+        A pure function of (param order, world, partition size,
+        compression config) — identical on every rank without
+        communication."""
+        dt, dev = self._grad_dt, self._device
+        part_elems = self._part_elems
 
         # Per-parameter compression overrides (reference: per-param
         # byteps_* attrs, mxnet/__init__.py:250-317): params whose
@@ -191,6 +240,12 @@ class GradEngine:
         import math
         align = 64 * self.world // math.gcd(64, self.world)
 
+        # key namespace: generation-suffixed after a re-bucket so PS
+        # servers allocate fresh state (old nelem would be rejected by
+        # the server's re-init validation)
+        key_fmt = "byteps.Partition.%d" if self._generation == 0 \
+            else "byteps.Partition.g%d.%%d" % self._generation
+
         self.buckets: List[Bucket] = []
         self.param_bucket: Dict[int, List[Bucket]] = {}   # param idx → buckets
         for gkey, idxs in groups.items():
@@ -201,8 +256,7 @@ class GradEngine:
                 index = len(self.buckets)
                 bucket = Bucket(plan=plan, buffer=buf, params=[], grads=[])
                 bucket.plan.index = index
-                key = C._state.registry.declare(
-                    "byteps.Partition.%d" % index)
+                key = C._state.registry.declare(key_fmt % index)
                 bucket.declared_key = key
                 bucket.compression_params = group_cfg[gkey]
                 for span in plan.spans:
@@ -242,38 +296,28 @@ class GradEngine:
                 p.grad = torch.zeros_like(p)
                 self._split_params[pidx] = bks
 
-        # optional reduced-precision wire for the RCCL collectives
-        # (``comm_dtype`` argument — set by Compression.fp16/bf16 — or
-        # BPS_COMM_DTYPE=bf16): cast bucket → persistent 16-bit scratch,
-        # all-reduce the scratch, cast back at synchronize.  Halves xGMI
-        # bytes; gradients still ACCUMULATE at full precision — only the
-        # wire is narrowed, matching the reference's Compression semantics
-        # (torch/compression.py:34-76).
-        if comm_dtype is None:
-            from ..common.config import env_str
-            wire = env_str("BPS_COMM_DTYPE", default="").lower()
-            comm_dtype = torch.bfloat16 if wire in ("bf16", "bfloat16") \
-                else torch.float16 if wire in ("fp16", "half") else None
-        self.comm_dtype = comm_dtype
-        if self.comm_dtype is not None and dt == self.comm_dtype:
-            self.comm_dtype = None
-        self._wire_scratch: Dict[int, torch.Tensor] = {}
-        if self.comm_dtype is not None:
-            # eager: the fused cast+scale descriptor needs stable pointers
+        # per-bucket comm scratch invalidated by any rebuild
+        self._wire_scratch = {}
+        if getattr(self, "comm_dtype", None) is not None:
             for b in self.buckets:
                 self._wire_scratch[b.plan.index] = torch.empty_like(
                     b.buffer, dtype=self.comm_dtype)
-        self._fused_desc = None     # (desc dev tensor, total_vec, vec)
+        self._fused_desc = None
 
-        self._ps = None
-        if C._state.ps_enabled:
-            from . import ps_pipeline
-            self._ps = ps_pipeline.get_pipeline(self)
-
-        self._attach_hooks()
-        log.debug("GradEngine: %d params → %d buckets (%.1f MiB each max)",
-                  len(self.params), len(self.buckets),
-                  part_elems * buf.element_size() / 2**20)
+    def rebucket(self) -> None:
+        """Elastic resume with a changed world: rebuild the bucket plan
+        under the new alignment and re-alias every ``p.grad`` (the
+        reference re-declared keys in original order,
+        common/operations.cc:96-119; here the plan itself depends on
+        world, so it is rebuilt deterministically on every rank)."""
+        self._generation += 1
+        with self._lock:
+            self._pending.clear()
+            self._inflight_bytes = 0
+            self._ready_params = 0
+        self._build_buckets()
+        log.info("rebucket: generation %d, %d buckets for world %d",
+                 self._generation, len(self.buckets), self.world)
 
     # -- helpers -----------------------------------------------------------
 
@@ -529,21 +573,15 @@ def _suspend_engines() -> None:
 def _resume_engines() -> None:
     for e in _engines:
         e._attach_hooks()
-        # topology may have changed — refresh the averaging divisor.
-        # (Bucket layout is NOT rebuilt: if the new world no longer
-        # divides the bucket alignment the PS shard path would truncate,
-        # so warn loudly — full elastic re-bucketing is future work; the
-        # reference likewise only re-declared keys, operations.cc:96-119.)
+        # topology may have changed: rebuild the bucket plan for the new
+        # world (alignment and PS shard math are world-dependent).  Keys
+        # move to a generation-suffixed namespace so PS servers allocate
+        # fresh state (reference re-declared keys in original order,
+        # operations.cc:96-119).
         new_world = dist.get_world_size() if dist.is_initialized() else 1
         if new_world != e.world:
             e.world = new_world
-            for b in e.buckets:
-                if b.buffer.numel() % max(1, new_world) != 0:
-                    log.warning(
-                        "resume: bucket %d (%d elems) not divisible by new "
-                        "world %d — PS sharding disabled would be unsafe; "
-                        "recreate the optimizer/DDP for clean re-bucketing",
-                        b.plan.index, b.buffer.numel(), new_world)
+            e.rebucket()
         e._ps = None
         if C._state.ps_enabled:
             from . import ps_pipeline

@@ -98,3 +98,95 @@ def test_launcher_cpu_ranges():
     r = cpu_ranges(2)
     assert len(r) == 2
     assert set(r[0]).isdisjoint(r[1]) or len(os.sched_getaffinity(0)) < 2
+
+
+# -- elastic re-bucketing: world 2 → 3 ---------------------------------------
+
+def _make_grow_model():
+    torch.manual_seed(7)
+    return torch.nn.Sequential(
+        torch.nn.Linear(23, 130), torch.nn.ReLU(), torch.nn.Linear(130, 5))
+
+
+def _grow_worker(rank, world, tmpdir):
+    import os
+    import time
+    import byteps_amd.torch as bps
+
+    base_port = int(os.environ["MASTER_PORT"])
+    m = _make_grow_model()
+    torch.manual_seed(42)
+    xs = [torch.randn(4, 23) for _ in range(3)]
+    ys = [torch.randn(4, 5) for _ in range(3)]
+
+    def mark(tag):
+        open(os.path.join(tmpdir, "%s.%d" % (tag, rank)), "w").close()
+
+    def wait(tag, ranks):
+        while not all(os.path.exists(os.path.join(tmpdir, "%s.%d" % (tag, r)))
+                      for r in ranks):
+            time.sleep(0.05)
+
+    if rank < 2:
+        os.environ["WORLD_SIZE"] = "2"
+        bps.init()
+        opt = bps.DistributedOptimizer(
+            torch.optim.SGD(m.parameters(), lr=0.05),
+            named_parameters=m.named_parameters())
+        for _ in range(2):
+            opt.zero_grad()
+            ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+            opt.step()
+        bps.suspend()
+        mark("suspended")
+        wait("suspended", (0, 1))
+        os.environ["WORLD_SIZE"] = "3"
+        bps.resume(num_workers=3, num_servers=0)
+    else:
+        wait("suspended", (0, 1))
+        os.environ["WORLD_SIZE"] = "3"
+        os.environ["MASTER_PORT"] = str(base_port + 1)  # resume's rotation
+        bps.init()
+        opt = bps.DistributedOptimizer(
+            torch.optim.SGD(m.parameters(), lr=0.05),
+            named_parameters=m.named_parameters())
+
+    # elastic join protocol: sync params from rank 0 (reference
+    # broadcast_parameters at (re)start, torch/__init__.py:268-299)
+    bps.broadcast_parameters(m.state_dict(), root_rank=0)
+    # one synced step at world 3: every bucket must divide by 3 and the
+    # averaged grads must match the 3-way big batch
+    opt.zero_grad()
+    ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+    opt.synchronize()
+    eng = opt._engine
+    assert eng.world == 3
+    for b in eng.buckets:
+        assert b.buffer.numel() % 3 == 0
+    grads = [p.grad.detach().clone() for p in m.parameters()]
+    params = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return grads, params
+
+
+def test_elastic_rebucket_grow_2_to_3(tmp_path):
+    results = run_in_processes(_grow_worker, 3, str(tmp_path),
+                               extra_env={"BPS_PARTITION_BYTES": "16384"})
+    # all ranks agree bit-for-bit after the world-3 step
+    for got, _ in results[1:]:
+        for a, b in zip(results[0][0], got):
+            assert torch.equal(a, b), "ranks disagree after re-bucketing"
+    # and the averaged grad equals the big-batch grad on the broadcast
+    # (rank-0) params
+    m = _make_grow_model()
+    with torch.no_grad():
+        for p, v in zip(m.parameters(), results[0][1]):
+            p.copy_(v)
+    torch.manual_seed(42)
+    xs = [torch.randn(4, 23) for _ in range(3)]
+    ys = [torch.randn(4, 5) for _ in range(3)]
+    loss = sum(((m(x) - y) ** 2).mean() for x, y in zip(xs, ys)) / 3
+    loss.backward()
+    for got, exp in zip(results[0][0], [p.grad for p in m.parameters()]):
+        assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6), \
+            "averaged grads diverge from the 3-way big batch"
