@@ -21,6 +21,8 @@
 
 #include <arpa/inet.h>
 #include <fcntl.h>
+#include <sched.h>
+#include <cstdio>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
 #include <sys/mman.h>
@@ -217,6 +219,47 @@ struct KeyState {
   std::vector<float> ef_err;     // residual of the previous reply
   std::vector<float> ef_comp;    // compensated merge workspace
 };
+
+// NUMA topology from sysfs (north star: "NUMA-aware CPU server" —
+// reference used libnuma-bound shm per PCIe switch, shared_memory.cc:52-82
+// and numa_bind, global.cc:194-205.  Here: engine threads are pinned
+// round-robin across nodes and each key's store is first-touched by its
+// engine thread, so merge traffic stays node-local).
+std::vector<std::vector<int>> numa_cpu_nodes() {
+  std::vector<std::vector<int>> nodes;
+  for (int nid = 0;; ++nid) {
+    char path[96];
+    snprintf(path, sizeof(path), "/sys/devices/system/node/node%d/cpulist",
+             nid);
+    FILE* f = fopen(path, "r");
+    if (!f) break;
+    char buf[4096];
+    std::vector<int> cpus;
+    if (fgets(buf, sizeof(buf), f)) {
+      // parse "0-7,16-23" style ranges
+      char* save = nullptr;
+      for (char* tok = strtok_r(buf, ",\n", &save); tok;
+           tok = strtok_r(nullptr, ",\n", &save)) {
+        int lo, hi;
+        if (sscanf(tok, "%d-%d", &lo, &hi) == 2)
+          for (int c = lo; c <= hi; ++c) cpus.push_back(c);
+        else if (sscanf(tok, "%d", &lo) == 1)
+          cpus.push_back(lo);
+      }
+    }
+    fclose(f);
+    if (!cpus.empty()) nodes.push_back(std::move(cpus));
+  }
+  return nodes;
+}
+
+void pin_to_cpus(const std::vector<int>& cpus) {
+  if (cpus.empty()) return;
+  cpu_set_t set;
+  CPU_ZERO(&set);
+  for (int c : cpus) CPU_SET(c, &set);
+  sched_setaffinity(0, sizeof(set), &set);
+}
 
 struct Task {
   std::shared_ptr<Conn> conn;
@@ -498,9 +541,11 @@ class Server {
         slot->levels = std::max(1u, ip.levels);
         slot->async_mode = cmd_async(h.cmd);
         slot->server_ef = (ip.flags & 1u) != 0;
-        // first-init page-aligned store (reference
-        // server/server.cc:266-294); vector is 64 B aligned via resize
-        slot->store.assign(ip.nelem, 0.0f);
+        // store allocation is DEFERRED to the first push so the
+        // engine thread that owns this key first-touches the pages on
+        // its own NUMA node (reference pre-allocated page-aligned at
+        // init, server/server.cc:266-294 — NUMA placement came from
+        // numactl there)
         // engine assignment by least accumulated load (reference
         // server/server.h:154-178)
         int best = 0;
@@ -529,6 +574,8 @@ class Server {
   void handle_pull(const std::shared_ptr<Conn>& conn, const MsgHeader& h,
                    KeyState* ks, bool ipc, const IpcExt& ext) {
     std::unique_lock<std::mutex> lk(ks->mu);
+    if (ks->store.size() != ks->nelem)
+      ks->store.assign(ks->nelem, 0.0f);   // pull-before-push: zeros
     uint64_t want_version = h.aux;
     PendingPull p{conn, h, ipc, ext};
     if (ks->async_mode || ks->version >= want_version) {
@@ -574,6 +621,14 @@ class Server {
   }
 
   void engine_loop(int tid) {
+    // NUMA: pin this engine (and the OMP team it spawns — workers
+    // inherit the creating thread's mask) to one node, round-robin.
+    // Single-node boxes: no-op.  BPS_SERVER_NUMA=0 disables.
+    static std::vector<std::vector<int>> nodes = numa_cpu_nodes();
+    const char* nenv = getenv("BPS_SERVER_NUMA");
+    bool numa_on = !(nenv && nenv[0] == '0');
+    if (numa_on && nodes.size() > 1)
+      pin_to_cpus(nodes[tid % nodes.size()]);
 #ifdef _OPENMP
     // Cap this engine thread's OMP team: with N engine threads each
     // spawning a default (all-cores) team, the server oversubscribes
@@ -599,6 +654,8 @@ class Server {
     const uint32_t codec = cmd_codec(t.hdr.cmd);
     const bool first = ks->async_mode ? false : ks->round_senders.empty();
     const int64_t n = (int64_t)ks->nelem;
+    if (ks->store.size() != ks->nelem)
+      ks->store.assign(ks->nelem, 0.0f);   // first-touch on THIS node
     float* acc = ks->store.data();
 
     switch (codec) {
