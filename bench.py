@@ -141,8 +141,14 @@ def main():
     cparams = None
     if args.compression != "none":
         cparams = {"compressor_type": args.compression,
-                   "ef_type": "vanilla",
-                   "compressor_k": 4096}
+                   "ef_type": "vanilla"}
+        if args.compression in ("topk", "randomk"):
+            cparams["compressor_k"] = 4096
+        elif args.compression == "dithering":
+            # natural s-level partitions (s ≤ 127; small s → sparse
+            # Elias wire)
+            cparams["compressor_k"] = 8
+            cparams["partition"] = "natural"
     from byteps_amd.torch.parallel import DistributedDataParallel as DDP
     model = DDP(net, broadcast_buffers=False, compression_params=cparams)
     opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
