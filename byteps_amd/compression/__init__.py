@@ -227,6 +227,7 @@ class NesterovMomentum(BaseCompressor):
 
     def __init__(self, inner: BaseCompressor, mu: float = 0.9):
         self.inner = inner
+        self.host_wire = getattr(inner, "host_wire", False)
         self.mu = mu
         self.codec = inner.codec
         self.levels = inner.levels
@@ -242,6 +243,12 @@ class NesterovMomentum(BaseCompressor):
     def decompress(self, payload, n, aux=0, out=None):
         return self.inner.decompress(payload, n, aux, out)
 
+    def encode_wire(self, host_payload, n, out_buf):
+        return self.inner.encode_wire(host_payload, n, out_buf)
+
+    def decode_wire(self, wire, n):
+        return self.inner.decode_wire(wire, n)
+
     def update_error(self, x, comp, err):
         self.inner.update_error(x, comp, err)
 
@@ -254,6 +261,7 @@ class ErrorFeedback(BaseCompressor):
 
     def __init__(self, inner: BaseCompressor):
         self.inner = inner
+        self.host_wire = getattr(inner, "host_wire", False)
         self.codec = inner.codec
         self.levels = inner.levels
         self._err: Optional[torch.Tensor] = None
@@ -269,6 +277,12 @@ class ErrorFeedback(BaseCompressor):
 
     def decompress(self, payload, n, aux=0, out=None):
         return self.inner.decompress(payload, n, aux, out)
+
+    def encode_wire(self, host_payload, n, out_buf):
+        return self.inner.encode_wire(host_payload, n, out_buf)
+
+    def decode_wire(self, wire, n):
+        return self.inner.decode_wire(wire, n)
 
 
 _REGISTRY = {

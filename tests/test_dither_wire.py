@@ -150,6 +150,7 @@ def _dither_ps_worker(rank, world, sparse):
         named_parameters=m.named_parameters(),
         compression_params={"compressor_type": "dithering",
                             "partition": "natural",
+                            "ef_type": "vanilla",
                             "compressor_k": 4})
     torch.manual_seed(42)
     x = torch.randn(8, 64)
