@@ -13,7 +13,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "_core.so")
 
-SOURCES = ["core.cc", "kv.cc", "server.cc", "cpu_reducer.cc",
+SOURCES = ["core.cc", "kv.cc", "server.cc", "cpu_reducer.cc", "rdma.cc",
            "kernels.hip", "compress.hip", "bn.hip", "ln.hip"]
 
 

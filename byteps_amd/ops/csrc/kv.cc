@@ -38,6 +38,7 @@
 #include <pybind11/stl.h>
 
 #include "kv.h"
+#include "rdma_abi.h"
 
 namespace py = pybind11;
 
@@ -120,6 +121,7 @@ struct Region {
   uint64_t size = 0;
   uint64_t used = 0;
   bool hip_registered = false;
+  bpsrdma::ibv_mr* mr = nullptr;   // RDMA lane: registered MR
 };
 
 class ServerConn {
@@ -192,6 +194,15 @@ class ServerConn {
     }
     shm_unlink(nm);  // server holds its own mapping (or failed)
     if (aux == ~0ULL) {
+      // not colocated: try the RDMA lane — register this region as an
+      // MR and announce {addr, rkey} instead of a shm name (the
+      // register-once reuse of reference server/server.cc:39-80)
+      if (try_rdma_region(r)) {
+        uintptr_t p = (uintptr_t)r.base;
+        r.used = nbytes;
+        regions_.push_back(r);
+        return p;
+      }
       if (r.hip_registered) (void)hipHostUnregister(r.base);
       munmap(r.base, r.size);
       ipc_enabled_ = false;
@@ -201,6 +212,54 @@ class ServerConn {
     r.used = nbytes;
     regions_.push_back(r);
     return p;
+  }
+
+  // Bootstrap the RC QP (once) and announce a registered region.
+  // Returns false when RDMA is unavailable/denied → caller falls back.
+  bool try_rdma_region(Region& r) {
+    using namespace bpsrdma;
+    if (!rdma_available()) return false;
+    if (!rdma_) {
+      RdmaConn* c = rdma_conn_create();
+      if (!c) return false;
+      RdmaPeerInfo mine = rdma_conn_local_info(c);
+      RdmaPeerInfo peer{};
+      auto req = submit(kRdmaConnect, 0, &mine, sizeof(mine), &peer,
+                        sizeof(peer), 0, 0);
+      uint64_t aux = ~0ULL, rlen = 0;
+      {
+        std::unique_lock<std::mutex> rlk(req->mu);
+        req->cv.wait(rlk, [&] { return req->done; });
+        if (req->error.empty()) {
+          aux = req->reply_aux;
+          rlen = req->reply_len;
+        }
+      }
+      if (aux == ~0ULL || rlen != sizeof(peer) ||
+          !rdma_conn_connect(c, peer)) {
+        rdma_conn_destroy(c);
+        return false;
+      }
+      rdma_ = c;
+    }
+    ibv_mr* mr = rdma_conn_reg(rdma_, r.base, r.size);
+    if (!mr) return false;
+    RdmaRegionInfo ri{(uint64_t)(uintptr_t)r.base, r.size, mr->rkey, 0};
+    uint32_t region_id = (uint32_t)regions_.size();
+    auto req = submit(kRdmaHello, region_id, &ri, sizeof(ri), nullptr, 0,
+                      0, r.size);
+    uint64_t aux = ~0ULL;
+    {
+      std::unique_lock<std::mutex> rlk(req->mu);
+      req->cv.wait(rlk, [&] { return req->done; });
+      if (req->error.empty()) aux = req->reply_aux;
+    }
+    if (aux == ~0ULL) {
+      rdma_mr_dereg(mr);
+      return false;
+    }
+    r.mr = mr;
+    return true;
   }
 
   bool ipc_active() {
@@ -257,8 +316,7 @@ class ServerConn {
                                   uint32_t cmd, uint64_t aux) {
     auto req = std::make_shared<Request>();
     req->hdr = MsgHeader{kMagic, op, key,
-                         (op == kPush || op == kInit || op == kIpcHello)
-                             ? len : 0,
+                         (op == kPull || op == kBarrier) ? 0 : len,
                          aux, rank_, cmd, seq_.fetch_add(1)};
     req->send_payload = payload;
     req->recv_buf = recv_buf;
@@ -400,6 +458,7 @@ class ServerConn {
   std::atomic<uint64_t> seq_{1};
   std::atomic<bool> closed_{false};
   bool ipc_enabled_ = false;
+  bpsrdma::RdmaConn* rdma_ = nullptr;
   uint64_t default_region_ = 64ULL << 20;
   std::mutex regions_mu_;
   std::vector<Region> regions_;

@@ -24,6 +24,11 @@ enum MsgOp : uint32_t {
   kIpcHello = 10,   // announce a shm region: payload = shm name,
                     // aux = size, key = region id (colocated fast path)
   kIpcHelloReply = 11,
+  kRdmaConnect = 12,      // payload = RdmaPeerInfo; reply carries server's
+  kRdmaConnectReply = 13,
+  kRdmaHello = 14,        // announce a REGISTERED region: payload =
+                          // {u64 addr, u64 size, u32 rkey}, key = region id
+  kRdmaHelloReply = 15,
 };
 
 // Colocated IPC fast path: when cmd has kCmdIpcPayload set, the frame

@@ -55,6 +55,7 @@
 
 #include "common.h"
 #include "kv.h"
+#include "rdma_abi.h"
 
 namespace py = pybind11;
 
@@ -111,6 +112,12 @@ bool read_all_fd(int fd, void* buf, size_t n) {
   return true;
 }
 
+struct RdmaRemoteRegion {
+  uint64_t addr = 0;
+  uint64_t size = 0;
+  uint32_t rkey = 0;
+};
+
 struct Conn {
   int fd;
   std::mutex write_mu;
@@ -118,9 +125,55 @@ struct Conn {
   // reader thread installs, engine threads look up
   std::mutex regions_mu;
   std::unordered_map<uint32_t, std::pair<char*, size_t>> regions;
+  // RDMA lane (remote worker): RC endpoint + the worker's registered
+  // regions; bounce buffers are this server's registered staging
+  bpsrdma::RdmaConn* rdma = nullptr;
+  std::unordered_map<uint32_t, RdmaRemoteRegion> rdma_regions;
+  char* rd_bounce = nullptr;       // reader-thread READ target
+  size_t rd_cap = 0;
+  bpsrdma::ibv_mr* rd_mr = nullptr;
+  char* wr_bounce = nullptr;       // reply WRITE source (under write_mu)
+  size_t wr_cap = 0;
+  bpsrdma::ibv_mr* wr_mr = nullptr;
 
   ~Conn() {
     for (auto& kv : regions) munmap(kv.second.first, kv.second.second);
+    if (rd_mr) bpsrdma::rdma_mr_dereg(rd_mr);
+    if (wr_mr) bpsrdma::rdma_mr_dereg(wr_mr);
+    free(rd_bounce);
+    free(wr_bounce);
+    if (rdma) bpsrdma::rdma_conn_destroy(rdma);
+  }
+
+  bool rdma_lookup(uint64_t locator, uint64_t len, uint64_t* raddr,
+                   uint32_t* rkey) {
+    std::lock_guard<std::mutex> lk(regions_mu);
+    auto it = rdma_regions.find(bpsamd::locator_region(locator));
+    if (it == rdma_regions.end()) return false;
+    uint64_t off = bpsamd::locator_off(locator);
+    if (off + len > it->second.size) return false;
+    *raddr = it->second.addr + off;
+    *rkey = it->second.rkey;
+    return true;
+  }
+
+  bool ensure_bounce(char** buf, size_t* cap, bpsrdma::ibv_mr** mr,
+                     size_t need) {
+    if (*cap >= need) return true;
+    size_t ncap = need < (8u << 20) ? (8u << 20) : need;
+    char* nb = (char*)aligned_alloc(4096, (ncap + 4095) & ~4095UL);
+    if (!nb) return false;
+    bpsrdma::ibv_mr* nmr = bpsrdma::rdma_conn_reg(rdma, nb, ncap);
+    if (!nmr) {
+      free(nb);
+      return false;
+    }
+    if (*mr) bpsrdma::rdma_mr_dereg(*mr);
+    free(*buf);
+    *buf = nb;
+    *cap = ncap;
+    *mr = nmr;
+    return true;
   }
 
   char* resolve(uint64_t locator, uint64_t len) {
@@ -419,6 +472,36 @@ class Server {
         if (!read_all_fd(conn->fd, &ext, sizeof(ext))) return;
         if (h.op == kPush) {
           ipc_ptr = conn->resolve(ext.locator, h.len);
+          if (!ipc_ptr && conn->rdma) {
+            // RDMA lane: pull the payload with a one-sided READ from
+            // the worker's registered staging into this reader's
+            // bounce, then process inline (the bounce is reused per
+            // frame, so the merge must complete before the next read)
+            uint64_t raddr;
+            uint32_t rkey;
+            if (conn->rdma_lookup(ext.locator, h.len, &raddr, &rkey) &&
+                conn->ensure_bounce(&conn->rd_bounce, &conn->rd_cap,
+                                    &conn->rd_mr, h.len) &&
+                bpsrdma::rdma_conn_read(conn->rdma, conn->rd_bounce,
+                                        conn->rd_mr->lkey, raddr, rkey,
+                                        (uint32_t)h.len)) {
+              KeyState* ks2 = key_state(h.key);
+              if (!ks2) {
+                reply_err(conn, h);
+                continue;
+              }
+              {
+                std::lock_guard<std::mutex> lk(ks2->mu);
+                ks2->push_total++;
+              }
+              Task t2{conn, h, {}, ks2};
+              t2.ipc_payload = conn->rd_bounce;
+              process_push(t2);
+              continue;
+            }
+            reply_err(conn, h);
+            continue;
+          }
           if (!ipc_ptr) {
             reply_err(conn, h);
             continue;
@@ -432,6 +515,12 @@ class Server {
       switch (h.op) {
         case kIpcHello:
           handle_ipc_hello(conn, h, payload);
+          break;
+        case kRdmaConnect:
+          handle_rdma_connect(conn, h, payload);
+          break;
+        case kRdmaHello:
+          handle_rdma_hello(conn, h, payload);
           break;
         case kInit:
           handle_init(conn, h, payload);
@@ -565,6 +654,52 @@ class Server {
     conn->send(r, nullptr);
   }
 
+  void handle_rdma_connect(const std::shared_ptr<Conn>& conn,
+                           const MsgHeader& h,
+                           const std::vector<char>& payload) {
+    MsgHeader r = h;
+    r.op = kRdmaConnectReply;
+    r.len = 0;
+    r.aux = ~0ULL;
+    if (bpsrdma::rdma_available() &&
+        payload.size() >= sizeof(bpsrdma::RdmaPeerInfo) && !conn->rdma) {
+      bpsrdma::RdmaPeerInfo peer{};
+      std::memcpy(&peer, payload.data(), sizeof(peer));
+      bpsrdma::RdmaConn* c = bpsrdma::rdma_conn_create();
+      if (c) {
+        bpsrdma::RdmaPeerInfo mine = bpsrdma::rdma_conn_local_info(c);
+        if (bpsrdma::rdma_conn_connect(c, peer)) {
+          conn->rdma = c;
+          r.aux = 0;
+          r.len = sizeof(mine);
+          conn->send(r, &mine);
+          return;
+        }
+        bpsrdma::rdma_conn_destroy(c);
+      }
+    }
+    conn->send(r, nullptr);
+  }
+
+  void handle_rdma_hello(const std::shared_ptr<Conn>& conn,
+                         const MsgHeader& h,
+                         const std::vector<char>& payload) {
+    MsgHeader r = h;
+    r.op = kRdmaHelloReply;
+    r.len = 0;
+    if (conn->rdma && payload.size() >= sizeof(bpsrdma::RdmaRegionInfo)) {
+      bpsrdma::RdmaRegionInfo ri{};
+      std::memcpy(&ri, payload.data(), sizeof(ri));
+      std::lock_guard<std::mutex> lk(conn->regions_mu);
+      conn->rdma_regions[(uint32_t)h.key] =
+          RdmaRemoteRegion{ri.addr, ri.size, ri.rkey};
+      r.aux = h.key;
+    } else {
+      r.aux = ~0ULL;
+    }
+    conn->send(r, nullptr);
+  }
+
   int engine_of(uint64_t key) {
     std::lock_guard<std::mutex> lk(keys_mu_);
     auto it = engine_of_.find(key);
@@ -612,6 +747,42 @@ class Server {
         std::memcpy(dst, src, len);
         p.conn->send_ext(r, p.ext);
         return;
+      }
+      if (p.conn->rdma && len <= p.ext.cap) {
+        // remote worker over verbs: stage into the registered reply
+        // bounce and RDMA WRITE it into the worker's recv staging,
+        // then ship the header (write_mu serializes bounce use)
+        uint64_t raddr;
+        uint32_t rkey;
+        if (p.conn->rdma_lookup(p.ext.locator, len, &raddr, &rkey)) {
+          std::lock_guard<std::mutex> wlk(p.conn->write_mu);
+          if (p.conn->ensure_bounce(&p.conn->wr_bounce, &p.conn->wr_cap,
+                                    &p.conn->wr_mr, len)) {
+            std::memcpy(p.conn->wr_bounce, src, len);
+            if (bpsrdma::rdma_conn_write(p.conn->rdma, p.conn->wr_bounce,
+                                         p.conn->wr_mr->lkey, raddr, rkey,
+                                         (uint32_t)len)) {
+              // header without payload; ext echoed (mutex already held
+              // → inline the send)
+              char frame[sizeof(MsgHeader) + sizeof(IpcExt)];
+              std::memcpy(frame, &r, sizeof(MsgHeader));
+              std::memcpy(frame + sizeof(MsgHeader), &p.ext,
+                          sizeof(IpcExt));
+              const char* q = frame;
+              size_t left = sizeof(frame);
+              while (left > 0) {
+                ssize_t w = ::write(p.conn->fd, q, left);
+                if (w < 0) {
+                  if (errno == EINTR) continue;
+                  return;
+                }
+                q += w;
+                left -= (size_t)w;
+              }
+              return;
+            }
+          }
+        }
       }
       // capacity/region mismatch: fall back to inline (clears the IPC
       // bit so the client reads the payload from the socket)
