@@ -103,3 +103,48 @@ def test_cross_barrier_adam_world2():
     for got in results:
         for p_got, p_exp in zip(got, expected):
             assert torch.allclose(p_got, p_exp, rtol=1e-5, atol=1e-6)
+
+
+# -- CrossBarrier + PS mode (ordered all-gather issuer, ADVICE A4) -----------
+
+def _cb_ps_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    from byteps_amd.torch.cross_barrier import CrossBarrier
+    bps.init()
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05, momentum=0.9)
+    cb = CrossBarrier(m, opt)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    x, y = xs[rank], ys[rank]
+    for _ in range(steps):
+        cb.zero_grad()
+        ((m(x) - y) ** 2).mean().backward()
+        cb.step()
+    cb.synchronize()
+    out = [p.detach().clone() for p in m.parameters()]
+    cb.stop()
+    bps.shutdown()
+    return out
+
+
+def test_cross_barrier_ps_world2():
+    """CrossBarrier over the PS pipeline: the poller consumes buckets in
+    completion order while the issuer serializes the trailing
+    all-gathers — must match synchronous SGD (the unordered version
+    could deadlock/corrupt; ADVICE.md finding 4)."""
+    from byteps_amd.ops import _core
+    srv = _core.Server(0, 2, False)
+    srv.start()
+    try:
+        env = {"BPS_FORCE_DISTRIBUTED": "1",
+               "BPS_SERVER_URIS": "127.0.0.1:%d" % srv.port,
+               "BPS_NUM_SERVER": "1"}
+        expected = _baseline(2, 4)
+        results = run_in_processes(_cb_ps_worker, 2, 4, extra_env=env)
+        for r in range(2):
+            for got, exp in zip(results[r], expected):
+                assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5)
+    finally:
+        srv.stop()
