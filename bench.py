@@ -123,7 +123,9 @@ def main():
         os.environ.setdefault("BPS_MIN_COMPRESS_BYTES", "65536")
         if int(os.environ.get("LOCAL_RANK", "0")) == 0:
             from byteps_amd.ops import core
-            server = core().Server(ps_port, 8, False)
+            server = core().Server(
+                ps_port, 8,
+                os.environ.get("BPS_SERVER_ENABLE_SCHEDULE", "0") == "1")
             server.start()
 
     import byteps_amd.torch as bps
