@@ -9,6 +9,7 @@
 #include <algorithm>
 #include <cmath>
 #include <cstdint>
+#include <cstdlib>
 #include <cstring>
 #include <atomic>
 #include <vector>
@@ -318,6 +319,19 @@ namespace {
 
 constexpr int64_t kChunkElems = 1 << 16;
 
+// Concurrent pool threads each spawn their own OMP team; uncapped teams
+// (= all cores) from 7+ threads oversubscribe catastrophically on
+// 256-core boxes (same failure mode as the server engines, measured in
+// profiles/MEASUREMENTS.md).  Cap the wire codec's teams.
+inline int wire_threads() {
+  static int n = [] {
+    const char* e = getenv("BPS_WIRE_THREADS");
+    int v = e ? atoi(e) : 4;
+    return v < 1 ? 1 : v;
+  }();
+  return n;
+}
+
 struct BitWriter {
   uint8_t* p;
   uint64_t bitpos = 0;
@@ -396,7 +410,7 @@ int bps_cpu_dither_encode(const int8_t* code, int64_t n, uint8_t* out,
   std::vector<std::vector<uint8_t>> bufs(nchunks);
   std::vector<int64_t> sizes(nchunks, 0);
   bool overflow = false;
-#pragma omp parallel for schedule(dynamic)
+#pragma omp parallel for schedule(dynamic) num_threads(wire_threads())
   for (int64_t c = 0; c < nchunks; ++c) {
     int64_t lo = c * kChunkElems;
     int64_t len = std::min<int64_t>(kChunkElems, n - lo);
@@ -441,7 +455,7 @@ int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
   if (offs[nc] > in_len) return -1;
   std::memset(code, 0, (size_t)n);
   std::atomic<int> bad{0};
-#pragma omp parallel for schedule(dynamic)
+#pragma omp parallel for schedule(dynamic) num_threads(wire_threads())
   for (int64_t c = 0; c < (int64_t)nc; ++c) {
     int64_t lo = (int64_t)c * ce;
     int64_t len = std::min<int64_t>(ce, n - lo);
