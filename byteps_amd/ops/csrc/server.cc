@@ -923,16 +923,20 @@ class Server {
         if (p.hdr.aux <= ks->version) send_pull_reply(p, ks);
         else ks->pending.push_back(p);
       }
-      if (ks->server_ef && ks->codec == kOnebit) {
+      if (ks->server_ef && ks->codec != kRaw) {
         // deferred EF residual: only needed by NEXT round's pack, so it
         // runs after the queued pulls were answered (off the pull
         // critical path)
-        int64_t nn = (int64_t)ks->nelem;
-        int64_t nwords = (nn + 63) >> 6;
-        float sc;
-        std::memcpy(&sc, ks->reply.data() + nwords * 8, 4);
-        bps_cpu_onebit_err_update(ks->ef_comp.data(), nn, sc,
-                                  ks->ef_err.data());
+        if (ks->codec == kOnebit) {
+          int64_t nn = (int64_t)ks->nelem;
+          int64_t nwords = (nn + 63) >> 6;
+          float sc;
+          std::memcpy(&sc, ks->reply.data() + nwords * 8, 4);
+          bps_cpu_onebit_err_update(ks->ef_comp.data(), nn, sc,
+                                    ks->ef_err.data());
+        } else {
+          deferred_generic_ef(ks);
+        }
       }
     }
   }
@@ -1035,7 +1039,14 @@ class Server {
         break;
     }
 
-    if (ks->server_ef) {
+  }
+
+  // EF residual for the generic (non-onebit) codecs — runs AFTER the
+  // queued pulls flushed (only the NEXT round reads ef_err)
+  void deferred_generic_ef(KeyState* ks) {
+    const int64_t n = (int64_t)ks->nelem;
+    const float* acc = ks->ef_comp.data();
+    {
       // new residual = compensated merge − decompress(reply)
       ks->scratch.assign(n, 0.0f);
       float* dec = ks->scratch.data();
@@ -1081,6 +1092,7 @@ class Server {
       for (int64_t i = 0; i < n; ++i) err[i] = acc[i] - dec[i];
     }
   }
+
 
   void handle_barrier(const std::shared_ptr<Conn>& conn, const MsgHeader& h) {
     std::lock_guard<std::mutex> lk(barrier_mu_);
