@@ -477,6 +477,88 @@ int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
   return bad ? -1 : 0;
 }
 
+// Server reply path for dithering, two passes instead of
+// compensate/norm/quantize three:
+//   pass 1: comp = acc + err, accumulating the norm (L2 for natural,
+//           max for linear)
+//   pass 2: quantize comp with a CHEAP splitmix draw — the reply's
+//           stochastic rounding needs no GPU-twin reproducibility (codes
+//           ship on the wire; only the worker-side compress has a HIP
+//           twin), and the 3-hash uniform_at was the reply's hot loop.
+int bps_cpu_dither_compensate_norm(const float* acc, const float* err,
+                                   int64_t n, int natural, float* comp,
+                                   float* out_norm) {
+  if (natural) {
+    double ss = 0.0;
+#pragma omp parallel for reduction(+ : ss)
+    for (int64_t i = 0; i < n; ++i) {
+      float c = err ? acc[i] + err[i] : acc[i];
+      comp[i] = c;
+      ss += (double)c * c;
+    }
+    *out_norm = (float)std::sqrt(ss);
+  } else {
+    float mx = 0.0f;
+#pragma omp parallel for reduction(max : mx)
+    for (int64_t i = 0; i < n; ++i) {
+      float c = err ? acc[i] + err[i] : acc[i];
+      comp[i] = c;
+      float a = std::fabs(c);
+      if (a > mx) mx = a;
+    }
+    *out_norm = mx;
+  }
+  return 0;
+}
+
+static inline float cheap_uniform(uint64_t seed, uint64_t i) {
+  uint64_t z = seed + i * 0x9E3779B97F4A7C15ULL;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  z ^= z >> 27;
+  return (float)(z >> 40) * (1.0f / 16777216.0f);
+}
+
+int bps_cpu_dithering_compress_fast(const float* x, int64_t n, int s,
+                                    uint64_t seed, int natural, float norm,
+                                    int8_t* code) {
+#pragma omp parallel for
+  for (int64_t i = 0; i < n; ++i) {
+    float v = x[i];
+    float r = (norm > 0.0f) ? std::fabs(v) / norm : 0.0f;
+    if (!natural) {
+      float t = r * s;
+      int level = (int)t;
+      float frac = t - level;
+      level += (cheap_uniform(seed, (uint64_t)i) < frac) ? 1 : 0;
+      if (level > s) level = s;
+      code[i] = (int8_t)(v < 0.0f ? -level : level);
+    } else {
+      if (r <= 0.0f) {
+        code[i] = 0;
+        continue;
+      }
+      float lowest = std::ldexp(1.0f, 1 - s);
+      int ebits;
+      if (r < lowest) {
+        if (cheap_uniform(seed, (uint64_t)i) >= r / lowest) {
+          code[i] = 0;
+          continue;
+        }
+        ebits = 1 - s;
+      } else {
+        int e;
+        float m = std::frexp(r, &e);
+        float p_up = m * 2.0f - 1.0f;
+        ebits = e - 1 + ((cheap_uniform(seed, (uint64_t)i) < p_up) ? 1 : 0);
+        if (ebits > 0) ebits = 0;
+        if (ebits < 1 - s) ebits = 1 - s;
+      }
+      code[i] = (int8_t)(v < 0.0f ? -(ebits + s) : (ebits + s));
+    }
+  }
+  return 0;
+}
+
 // Fused dithering decode → accumulator (server push path, one pass over
 // dense codes)
 int bps_cpu_dithering_accumulate(const int8_t* code, int64_t n, int s,

@@ -85,6 +85,12 @@ int bps_cpu_dither_encode(const int8_t* code, int64_t n, uint8_t* out,
                           int64_t out_cap, int64_t* out_len);
 int bps_cpu_dither_decode(const uint8_t* in, int64_t in_len, int64_t n,
                           int8_t* code);
+int bps_cpu_dither_compensate_norm(const float* acc, const float* err,
+                                   int64_t n, int natural, float* comp,
+                                   float* out_norm);
+int bps_cpu_dithering_compress_fast(const float* x, int64_t n, int s,
+                                    uint64_t seed, int natural, float norm,
+                                    int8_t* code);
 int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
                         float* val);
 int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
@@ -963,7 +969,9 @@ class Server {
     // server/server.cc:228-257 + vanilla EF): compensate the merge with
     // the previous reply's residual before compressing, then store the
     // new residual.
-    if (ks->server_ef) {
+    const bool dither = ks->codec == kDitherLinear ||
+                        ks->codec == kDitherNatural;
+    if (ks->server_ef && !dither) {   // dithering fuses compensation
       if (ks->ef_err.empty()) ks->ef_err.assign(n, 0.0f);
       ks->ef_comp.resize(n);
       const float* err = ks->ef_err.data();
@@ -1007,12 +1015,29 @@ class Server {
       }
       case kDitherLinear:
       case kDitherNatural: {
+        // reply pack in two passes: fused compensate+norm, then a
+        // cheap-RNG quantize (cpu_reducer.cc) — EF compensation is
+        // folded here (acc already points at ef_comp when server_ef,
+        // but the fused version recomputes with the norm in one pass)
         bool natural = ks->codec == kDitherNatural;
-        float norm = bps_cpu_norm(acc, n, natural ? 1 : 2);
         uint64_t seed = splitmix64(ks->version * 0xD6E8FEB86659FD93ULL + 5);
+        float norm = 0.0f;
         ks->code_scratch.resize(n);
-        bps_cpu_dithering_compress(acc, n, (int)ks->levels, seed, natural,
-                                   norm, ks->code_scratch.data());
+        if (ks->server_ef) {
+          // fused compensate + norm (pass 1), cheap-RNG quantize (pass 2)
+          if (ks->ef_err.empty()) ks->ef_err.assign(n, 0.0f);
+          ks->ef_comp.resize(n);
+          bps_cpu_dither_compensate_norm(ks->store.data(),
+                                         ks->ef_err.data(), n,
+                                         natural ? 1 : 0,
+                                         ks->ef_comp.data(), &norm);
+          acc = ks->ef_comp.data();
+        } else {
+          norm = bps_cpu_norm(acc, n, natural ? 1 : 2);
+        }
+        bps_cpu_dithering_compress_fast(acc, n, (int)ks->levels, seed,
+                                        natural, norm,
+                                        ks->code_scratch.data());
         ks->reply.resize(5 + n);
         std::memcpy(ks->reply.data(), &norm, 4);
         int64_t wlen = 0;
