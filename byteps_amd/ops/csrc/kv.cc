@@ -304,10 +304,15 @@ class ServerConn {
     }
     std::lock_guard<std::mutex> rlk(regions_mu_);
     for (auto& r : regions_) {
+      if (r.mr) bpsrdma::rdma_mr_dereg(r.mr);
       if (r.hip_registered) (void)hipHostUnregister(r.base);
       munmap(r.base, r.size);
     }
     regions_.clear();
+    if (rdma_) {
+      bpsrdma::rdma_conn_destroy(rdma_);
+      rdma_ = nullptr;
+    }
   }
 
   std::shared_ptr<Request> submit(uint32_t op, uint64_t key,
