@@ -137,3 +137,29 @@ def test_ps_training_parity_ipc_vs_inline(ipc_on):
             assert torch.equal(a, b)
     finally:
         srv.stop()
+
+
+def test_ipc_region_growth(monkeypatch):
+    """Staging demand beyond one region must spawn more regions (each
+    with its own hello) and keep locators valid across all of them."""
+    monkeypatch.setenv("BPS_IPC_REGION_MB", "1")
+    from byteps_amd.ops import core
+    c = core()
+    srv = _server()
+    try:
+        kv = c.KVClient(0, ["127.0.0.1:%d" % srv.port])
+        import ctypes
+        bufs = []
+        # 6 x 512 KiB spans 3+ one-MiB regions
+        for i in range(6):
+            nb = 512 * 1024
+            addr = kv.ipc_alloc(0, nb)
+            assert addr != 0, "region growth failed at %d" % i
+            raw = (ctypes.c_uint8 * nb).from_address(addr)
+            bufs.append(torch.frombuffer(raw, dtype=torch.uint8))
+        # round-trip through buffers in the FIRST and LAST region
+        _roundtrip(kv, bufs[0], bufs[1], key=10, n=1024)
+        _roundtrip(kv, bufs[4], bufs[5], key=11, n=1024)
+        kv.close()
+    finally:
+        srv.stop()
