@@ -370,3 +370,77 @@ def test_ps_scheduler_discovery():
     finally:
         srv.stop()
         sched.stop()
+
+
+# -- multiple servers (BASELINE config 3 names 2 CPU PS) ---------------------
+
+def _make_wide_model(seed=0):
+    torch.manual_seed(seed)
+    # ~70k elems → ~16 buckets at the 4096-elem floor, so BOTH servers
+    # get keys
+    return torch.nn.Sequential(
+        torch.nn.Linear(64, 512), torch.nn.ReLU(),
+        torch.nn.Linear(512, 64))
+
+
+def _wide_baseline(world, steps, lr=0.05):
+    m = _make_wide_model()
+    opt = torch.optim.SGD(m.parameters(), lr=lr)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 64) for _ in range(world)]
+    ys = [torch.randn(8, 64) for _ in range(world)]
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = sum(((m(x) - y) ** 2).mean() for x, y in zip(xs, ys)) / world
+        loss.backward()
+        opt.step()
+    return [p.detach().clone() for p in m.parameters()]
+
+
+def _ps_multi_server_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_wide_model()
+    opt = bps.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05),
+        named_parameters=m.named_parameters())
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 64) for _ in range(world)]
+    ys = [torch.randn(8, 64) for _ in range(world)]
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        opt.step()
+    # both servers must actually own keys (djb2 spread)
+    import byteps_amd.common as C
+    loads = list(C._state.assigner.load)
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return loads, out
+
+
+def test_ps_two_servers_parity():
+    """2 workers + 2 PS shards: key→server hash sharding end-to-end
+    (reference mixed/colocated placement, common/global.cc:628-677)."""
+    from byteps_amd.ops import _core
+    s1 = _core.Server(0, 2, False)
+    s2 = _core.Server(0, 2, False)
+    s1.start()
+    s2.start()
+    try:
+        env = {"BPS_FORCE_DISTRIBUTED": "1",
+               "BPS_NUM_SERVER": "2",
+               "BPS_SERVER_URIS": "127.0.0.1:%d,127.0.0.1:%d"
+                                  % (s1.port, s2.port),
+               "BPS_PARTITION_BYTES": "4096"}
+        expected = _wide_baseline(2, 3)
+        results = run_in_processes(_ps_multi_server_worker, 2, 3,
+                                   extra_env=env)
+        for loads, params in results:
+            assert len(loads) == 2
+            assert min(loads) > 0, "one server owns no keys: %s" % loads
+            for got, exp in zip(params, expected):
+                assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5)
+    finally:
+        s1.stop()
+        s2.stop()
