@@ -581,7 +581,14 @@ def _resume_engines() -> None:
         new_world = dist.get_world_size() if dist.is_initialized() else 1
         if new_world != e.world:
             e.world = new_world
-            e.rebucket()
+            if e.on_bucket_issued is not None:
+                # a CrossBarrier owns per-bucket optimizers over the OLD
+                # plan — rebuilding under it would silently orphan them
+                log.warning(
+                    "resume: world changed under a CrossBarrier engine — "
+                    "recreate the CrossBarrier; skipping re-bucket")
+            else:
+                e.rebucket()
         e._ps = None
         if C._state.ps_enabled:
             from . import ps_pipeline
