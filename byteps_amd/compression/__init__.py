@@ -186,12 +186,20 @@ class DitheringCompressor(BaseCompressor):
         out_buf[4] = 1
         return 5 + int(wlen)
 
-    def decode_wire(self, wire: torch.Tensor, n: int) -> torch.Tensor:
-        """Sparse host wire → dense host payload (no-op for flag 0)."""
+    def decode_wire(self, wire: torch.Tensor, n: int,
+                    out=None) -> torch.Tensor:
+        """Sparse host wire → dense host payload (no-op for flag 0).
+        ``out`` (≥ 5+n bytes, ideally pinned — the pipeline reuses its
+        push wire buffer, free once the reply exists) avoids a pageable
+        H2D of the dense codes."""
         if int(wire[4]) != 1:
             return wire
         from ..ops import core
-        dense = torch.empty(5 + n, dtype=torch.uint8)
+        if out is not None and out.numel() >= 5 + n \
+                and out.data_ptr() != wire.data_ptr():
+            dense = out[:5 + n]
+        else:
+            dense = torch.empty(5 + n, dtype=torch.uint8)
         dense[:4] = wire[:4]
         dense[4] = 0
         body = wire[5:].contiguous()
@@ -246,8 +254,8 @@ class NesterovMomentum(BaseCompressor):
     def encode_wire(self, host_payload, n, out_buf):
         return self.inner.encode_wire(host_payload, n, out_buf)
 
-    def decode_wire(self, wire, n):
-        return self.inner.decode_wire(wire, n)
+    def decode_wire(self, wire, n, out=None):
+        return self.inner.decode_wire(wire, n, out)
 
     def update_error(self, x, comp, err):
         self.inner.update_error(x, comp, err)
@@ -281,8 +289,8 @@ class ErrorFeedback(BaseCompressor):
     def encode_wire(self, host_payload, n, out_buf):
         return self.inner.encode_wire(host_payload, n, out_buf)
 
-    def decode_wire(self, wire, n):
-        return self.inner.decode_wire(wire, n)
+    def decode_wire(self, wire, n, out=None):
+        return self.inner.decode_wire(wire, n, out)
 
 
 _REGISTRY = {

@@ -429,8 +429,9 @@ class PSPipeline:
             if comp is not None:
                 wire = st.recv[:reply_len]
                 if getattr(comp, "host_wire", False):
-                    # Elias reply → dense host payload before H2D
-                    wire = comp.decode_wire(wire, ki.nelem)
+                    # Elias reply → dense host payload before H2D; the
+                    # push wire buffer is pinned and free by reply time
+                    wire = comp.decode_wire(wire, ki.nelem, out=st.wire)
                 if on_gpu:
                     wire = wire.to(shard.device, non_blocking=True)
                 # sparse codecs: the server's reply k = min(levels, nelem);
