@@ -109,6 +109,19 @@ class SelfAttention(nn.Module):
         return self.out(o)
 
 
+def _use_fused_mlp() -> bool:
+    import os
+    if os.environ.get("BPS_FUSED_MLP", "1") in ("0", "false", "no"):
+        return False
+    if not torch.cuda.is_available():
+        return False
+    try:
+        from ..torch.fused_mlp import fused_mlp_available
+        return fused_mlp_available()
+    except Exception:
+        return False
+
+
 class EncoderLayer(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
@@ -117,10 +130,17 @@ class EncoderLayer(nn.Module):
         self.fc1 = nn.Linear(cfg.hidden, cfg.intermediate)
         self.fc2 = nn.Linear(cfg.intermediate, cfg.hidden)
         self.ln2 = FusedLayerNorm(cfg.hidden)
+        self._fused_mlp = _use_fused_mlp()
 
     def forward(self, x):
         x = self.ln1(x + self.attn(x))
-        h = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
+        if self._fused_mlp and x.is_cuda:
+            # GELU + both bias grads ride the hipBLASLt epilogues
+            from ..torch.fused_mlp import fused_mlp
+            h = fused_mlp(x, self.fc1.weight, self.fc1.bias,
+                          self.fc2.weight, self.fc2.bias)
+        else:
+            h = self.fc2(F.gelu(self.fc1(x), approximate="tanh"))
         return self.ln2(x + h)
 
 

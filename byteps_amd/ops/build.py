@@ -13,7 +13,7 @@ HERE = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(HERE, "csrc")
 OUT = os.path.join(HERE, "_core.so")
 
-SOURCES = ["core.cc", "kv.cc", "server.cc", "cpu_reducer.cc", "rdma.cc",
+SOURCES = ["core.cc", "kv.cc", "server.cc", "cpu_reducer.cc", "rdma.cc", "blaslt.cc",
            "kernels.hip", "compress.hip", "bn.hip", "ln.hip"]
 
 
@@ -36,6 +36,7 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> st
         hipcc, "--offload-arch=" + arch, "-O3", "-std=c++17", "-fPIC",
         "-shared", "-fopenmp", "-pthread", "-fvisibility=hidden",
         "-Wno-unused-result",
+        "-L/opt/rocm/lib", "-lhipblaslt",
         "-I", pybind11.get_include(),
         "-I", sysconfig.get_paths()["include"],
         *srcs,
