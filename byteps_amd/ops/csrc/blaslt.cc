@@ -180,7 +180,8 @@ int bps_lt_gemm_gelu_aux(const void* X, const void* W, const void* bias,
     lt_matmul(N, M, K, HIPBLAS_OP_T, HIPBLAS_OP_N, W, K, X, K, Y, N,
               HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, const_cast<void*>(bias),
               aux, N, (hipStream_t)stream);
-  } catch (const std::exception&) {
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[bps blaslt] %s\n", e.what());
     return -1;
   }
   return 0;
@@ -194,7 +195,8 @@ int bps_lt_gemm_bias(const void* X, const void* W, const void* bias,
     lt_matmul(N, M, K, HIPBLAS_OP_T, HIPBLAS_OP_N, W, K, X, K, Y, N,
               HIPBLASLT_EPILOGUE_BIAS, const_cast<void*>(bias), nullptr, 0,
               (hipStream_t)stream);
-  } catch (const std::exception&) {
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[bps blaslt] %s\n", e.what());
     return -1;
   }
   return 0;
@@ -208,7 +210,8 @@ int bps_lt_gemm_dgelu_bgrad(const void* dY2, const void* W2,
     lt_matmul(I, M, H, HIPBLAS_OP_N, HIPBLAS_OP_N, W2, I, dY2, H, dY1, I,
               HIPBLASLT_EPILOGUE_DGELU_BGRAD, db1,
               const_cast<void*>(aux), I, (hipStream_t)stream);
-  } catch (const std::exception&) {
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[bps blaslt] %s\n", e.what());
     return -1;
   }
   return 0;
@@ -222,7 +225,8 @@ int bps_lt_gemm_wgrad(const void* dY, const void* X, void* dW, void* db,
     lt_matmul(K, N, M, HIPBLAS_OP_N, HIPBLAS_OP_T, X, K, dY, N, dW, K,
               db ? HIPBLASLT_EPILOGUE_BGRADB : HIPBLASLT_EPILOGUE_DEFAULT,
               db, nullptr, 0, (hipStream_t)stream);
-  } catch (const std::exception&) {
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[bps blaslt] %s\n", e.what());
     return -1;
   }
   return 0;
@@ -235,7 +239,8 @@ int bps_lt_gemm_dgrad(const void* dY, const void* W, void* dX, int64_t M,
     lt_matmul(K, M, N, HIPBLAS_OP_N, HIPBLAS_OP_N, W, K, dY, N, dX, K,
               HIPBLASLT_EPILOGUE_DEFAULT, nullptr, nullptr, 0,
               (hipStream_t)stream);
-  } catch (const std::exception&) {
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[bps blaslt] %s\n", e.what());
     return -1;
   }
   return 0;
