@@ -5,13 +5,20 @@
 // (profiles/bert_large_final_kernels.txt).  hipBLASLt's epilogues fold
 // them into the GEMMs themselves (the idiomatic MFMA path — matrix-core
 // work stays in the library, pointwise work rides the epilogue):
-//   fwd1 : Y1  = GELU(X·W1ᵀ + b1),  aux = pre-GELU  (GELU_AUX_BIAS)
-//   fwd2 : Y2  = Y1·W2ᵀ + b2                        (BIAS)
-//   dgrad2: dY1 = dGELU(aux) ⊙ (dY2·W2), db1 = Σrows (DGELU_BGRAD)
-//   wgrad2: dW2 = dY2ᵀ·Y1, db2 = Σ       (BGRADB)
-//   wgrad1: dW1 = dY1ᵀ·X                            (DEFAULT)
-//   dgrad1: dX  = dY1·W1                            (DEFAULT)
+// Supported epilogue set on this hipBLASLt/gfx950 (probed,
+// scripts/probe_blaslt.cc): BIAS/GELU_BIAS all layouts, DGELU NN-only,
+// BGRADB NT-only — no AUX-output forward, no DGELU_BGRAD.  So the
+// forward stays eager (addmm+gelu keeps the pre-GELU H anyway) and the
+// BACKWARD is fully fused:
+//   fwd1 : H  = X·W1ᵀ + b1 (BIAS), Y1 = gelu(H) (elementwise)
+//   fwd2 : Y2 = Y1·W2ᵀ + b2 (BIAS)
+//   dgrad2: dY1 = dGELU(H) ⊙ (dY2·W2)            (DGELU, NN)
+//   wgrad2: dW2 = dY2ᵀ·Y1, db2 = Σ_M dY2         (BGRADB, NT)
+//   wgrad1: dW1 = dY1ᵀ·X,  db1 = Σ_M dY1         (BGRADB, NT)
+//   dgrad1: dX  = dY1·W1                         (DEFAULT)
 //
+// Bias vectors and bias-grad outputs are bf16 (= D type; matches eager
+// autocast, which also reduces bias grads from bf16).
 // Row-major torch tensors map onto hipBLASLt's column-major world by the
 // usual swap: C_row[M,N] = A_row[M,K]·B_row[K,N]  ⇔
 // C_col[N,M] = op(B_mem)·op(A_mem).  All matrices bf16, compute fp32,
@@ -104,17 +111,9 @@ void lt_matmul(int64_t m, int64_t n, int64_t k, hipblasOperation_t ta,
       hipblasLtEpilogue_t e = epi;
       BLT_CHECK(hipblasLtMatmulDescSetAttribute(
           plan->desc, HIPBLASLT_MATMUL_DESC_EPILOGUE, &e, sizeof(e)));
-      if (bias) {
-        int32_t btype = HIP_R_32F;
-        BLT_CHECK(hipblasLtMatmulDescSetAttribute(
-            plan->desc, HIPBLASLT_MATMUL_DESC_BIAS_DATA_TYPE, &btype,
-            sizeof(btype)));
-      }
+      // bias and aux stay at their defaults (= D type, bf16): algo
+      // coverage for non-default bias/aux dtypes is spotty
       if (aux) {
-        int32_t atype = HIP_R_16BF;
-        BLT_CHECK(hipblasLtMatmulDescSetAttribute(
-            plan->desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_DATA_TYPE,
-            &atype, sizeof(atype)));
         BLT_CHECK(hipblasLtMatmulDescSetAttribute(
             plan->desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_LD, &aux_ld,
             sizeof(aux_ld)));
@@ -172,21 +171,6 @@ void lt_matmul(int64_t m, int64_t n, int64_t k, hipblasOperation_t ta,
 // bias and bias-grads fp32.  S = hip stream.
 extern "C" {
 
-// Y[M,N] = GELU(X[M,K]·W[N,K]ᵀ + b[N]);  aux[M,N] bf16 = pre-GELU
-int bps_lt_gemm_gelu_aux(const void* X, const void* W, const void* bias,
-                         void* Y, void* aux, int64_t M, int64_t N,
-                         int64_t K, void* stream) {
-  try {
-    lt_matmul(N, M, K, HIPBLAS_OP_T, HIPBLAS_OP_N, W, K, X, K, Y, N,
-              HIPBLASLT_EPILOGUE_GELU_AUX_BIAS, const_cast<void*>(bias),
-              aux, N, (hipStream_t)stream);
-  } catch (const std::exception& e) {
-    fprintf(stderr, "[bps blaslt] %s\n", e.what());
-    return -1;
-  }
-  return 0;
-}
-
 // Y[M,N] = X[M,K]·W[N,K]ᵀ + b[N]
 int bps_lt_gemm_bias(const void* X, const void* W, const void* bias,
                      void* Y, int64_t M, int64_t N, int64_t K,
@@ -202,13 +186,13 @@ int bps_lt_gemm_bias(const void* X, const void* W, const void* bias,
   return 0;
 }
 
-// dY1[M,I] = dGELU(aux[M,I]) ⊙ (dY2[M,H]·W2[H,I]);  db1[I] fp32 = Σ_M dY1
-int bps_lt_gemm_dgelu_bgrad(const void* dY2, const void* W2,
-                            const void* aux, void* dY1, void* db1,
-                            int64_t M, int64_t H, int64_t I, void* stream) {
+// dY1[M,I] = dGELU(aux[M,I]) ⊙ (dY2[M,H]·W2[H,I])
+int bps_lt_gemm_dgelu(const void* dY2, const void* W2, const void* aux,
+                      void* dY1, int64_t M, int64_t H, int64_t I,
+                      void* stream) {
   try {
     lt_matmul(I, M, H, HIPBLAS_OP_N, HIPBLAS_OP_N, W2, I, dY2, H, dY1, I,
-              HIPBLASLT_EPILOGUE_DGELU_BGRAD, db1,
+              HIPBLASLT_EPILOGUE_DGELU, nullptr,
               const_cast<void*>(aux), I, (hipStream_t)stream);
   } catch (const std::exception& e) {
     fprintf(stderr, "[bps blaslt] %s\n", e.what());

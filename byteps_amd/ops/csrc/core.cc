@@ -24,15 +24,12 @@ int bps_cast_scale_many(const void* desc_dev, int nseg, int64_t total_vec,
                         void* stream);
 int bps_cast_scale(void* dst, const void* src, int64_t n, float alpha,
                    int src_dtype, int dst_dtype, void* stream);
-int bps_lt_gemm_gelu_aux(const void* X, const void* W, const void* bias,
-                         void* Y, void* aux, int64_t M, int64_t N,
-                         int64_t K, void* stream);
 int bps_lt_gemm_bias(const void* X, const void* W, const void* bias,
                      void* Y, int64_t M, int64_t N, int64_t K,
                      void* stream);
-int bps_lt_gemm_dgelu_bgrad(const void* dY2, const void* W2,
-                            const void* aux, void* dY1, void* db1,
-                            int64_t M, int64_t H, int64_t I, void* stream);
+int bps_lt_gemm_dgelu(const void* dY2, const void* W2, const void* aux,
+                      void* dY1, int64_t M, int64_t H, int64_t I,
+                      void* stream);
 int bps_lt_gemm_wgrad(const void* dY, const void* X, void* dW, void* db,
                       int64_t M, int64_t N, int64_t K, void* stream);
 int bps_lt_gemm_dgrad(const void* dY, const void* W, void* dX, int64_t M,
@@ -181,24 +178,17 @@ PYBIND11_MODULE(_core, m) {
         });
 
   // hipBLASLt epilogue-fused GEMMs (BERT MLP block)
-  m.def("lt_gemm_gelu_aux",
-        [](uintptr_t X, uintptr_t W, uintptr_t b, uintptr_t Y, uintptr_t aux,
-           int64_t M, int64_t N, int64_t K, uintptr_t s) {
-          check(bps_lt_gemm_gelu_aux(CP(X), CP(W), CP(b), P(Y), P(aux), M,
-                                     N, K, P(s)), "bps_lt_gemm_gelu_aux");
-        });
   m.def("lt_gemm_bias",
         [](uintptr_t X, uintptr_t W, uintptr_t b, uintptr_t Y, int64_t M,
            int64_t N, int64_t K, uintptr_t s) {
           check(bps_lt_gemm_bias(CP(X), CP(W), CP(b), P(Y), M, N, K, P(s)),
                 "bps_lt_gemm_bias");
         });
-  m.def("lt_gemm_dgelu_bgrad",
+  m.def("lt_gemm_dgelu",
         [](uintptr_t dY2, uintptr_t W2, uintptr_t aux, uintptr_t dY1,
-           uintptr_t db1, int64_t M, int64_t H, int64_t I, uintptr_t s) {
-          check(bps_lt_gemm_dgelu_bgrad(CP(dY2), CP(W2), CP(aux), P(dY1),
-                                        P(db1), M, H, I, P(s)),
-                "bps_lt_gemm_dgelu_bgrad");
+           int64_t M, int64_t H, int64_t I, uintptr_t s) {
+          check(bps_lt_gemm_dgelu(CP(dY2), CP(W2), CP(aux), P(dY1), M, H,
+                                  I, P(s)), "bps_lt_gemm_dgelu");
         });
   m.def("lt_gemm_wgrad",
         [](uintptr_t dY, uintptr_t X, uintptr_t dW, uintptr_t db, int64_t M,
