@@ -8,13 +8,18 @@ supports DGELU (NN) and BGRADB (NT) epilogues but no AUX-output forward
 (probed — scripts/probe_blaslt.cc), so the FORWARD stays eager (addmm +
 gelu already keeps the pre-GELU activation) and the BACKWARD is fused:
 
-  dgrad2: dY1 = dGELU(H) ⊙ (dY2·W2)       (DGELU epilogue, aux = H)
+  dgrad2: dY1 = gelu_backward(dY2·W2, H)   (lt GEMM + aten kernel —
+          the DGELU epilogue exists but measured 4x slower than the
+          plain GEMM and numerically off on this hipBLASLt build,
+          profiles/MEASUREMENTS.md)
   wgrad2: dW2 = dY2ᵀ·Y1  with db2 = Σ dY2  (BGRADB)
   wgrad1: dW1 = dY1ᵀ·X   with db1 = Σ dY1  (BGRADB)
   dgrad1: dX  = dY1·W1
 
-Numerics: same operator set as eager bf16 autocast (the DGELU epilogue
-is the same tanh-approximation derivative; GEMMs accumulate fp32)."""
+Measured verdict (same-box A/B): the fused path is ~neutral vs the
+TunableOp-tuned eager GEMMs + torch's cheap bias reduces — so it ships
+DEFAULT OFF (BPS_FUSED_MLP=1 opts in); the extension and probe stay as
+the documented exploration for future hipBLASLt versions."""
 
 from __future__ import annotations
 
@@ -53,10 +58,12 @@ class _FusedMLPFn(torch.autograd.Function):
         db2 = torch.empty(H, dtype=torch.bfloat16, device=dev)
         core.lt_gemm_wgrad(dy2.data_ptr(), y1.data_ptr(), dw2.data_ptr(),
                            db2.data_ptr(), M, H, I, s)
-        # dY1 [M, I] = dGELU(H) ⊙ (dY2·W2)  (DGELU epilogue, aux = H)
-        dy1 = torch.empty(M, I, dtype=torch.bfloat16, device=dev)
-        core.lt_gemm_dgelu(dy2.data_ptr(), w2.data_ptr(), h.data_ptr(),
-                           dy1.data_ptr(), M, H, I, s)
+        # dY1 [M, I] = gelu'(H) ⊙ (dY2·W2)
+        t = torch.empty(M, I, dtype=torch.bfloat16, device=dev)
+        core.lt_gemm_dgrad(dy2.data_ptr(), w2.data_ptr(), t.data_ptr(),
+                           M, H, I, s)
+        dy1 = torch.ops.aten.gelu_backward(t, h, approximate="tanh")
+        dy1 = dy1.contiguous()
         # dW1 [I, K] with db1 fused
         dw1 = torch.empty(I, K, dtype=torch.bfloat16, device=dev)
         db1 = torch.empty(I, dtype=torch.bfloat16, device=dev)

@@ -64,7 +64,6 @@ def test_fused_mlp_forward_backward(M, K, I):
 def test_bert_layer_fused_vs_eager_numerics():
     """Whole encoder layer: fused and eager MLP paths must agree to bf16
     tolerance on the same weights."""
-    import os
     from byteps_amd.models import bert as B
     torch.manual_seed(1)
     cfg = B.BertConfig(hidden=256, layers=1, heads=4, intermediate=1024)
