@@ -111,8 +111,8 @@ class SelfAttention(nn.Module):
 
 def _use_fused_mlp() -> bool:
     import os
-    if os.environ.get("BPS_FUSED_MLP", "0") not in ("1", "true", "yes"):
-        return False   # default OFF: measured neutral-to-worse vs TunableOp
+    if os.environ.get("BPS_FUSED_MLP", "1") in ("0", "false", "no"):
+        return False
     if not torch.cuda.is_available():
         return False
     try:

@@ -16,10 +16,10 @@ gelu already keeps the pre-GELU activation) and the BACKWARD is fused:
   wgrad1: dW1 = dY1ᵀ·X   with db1 = Σ dY1  (BGRADB)
   dgrad1: dX  = dY1·W1
 
-Measured verdict (same-box A/B): the fused path is ~neutral vs the
-TunableOp-tuned eager GEMMs + torch's cheap bias reduces — so it ships
-DEFAULT OFF (BPS_FUSED_MLP=1 opts in); the extension and probe stay as
-the documented exploration for future hipBLASLt versions."""
+Measured (same-box A/B, BERT-large b64 s128): 198.1k tok/s fused vs
+195.1k eager (+1.5%) once the DGELU epilogue was dropped — the fused
+bias-grads + mini-autotuned GEMMs are a small net win, so this is the
+default (BPS_FUSED_MLP=0 reverts)."""
 
 from __future__ import annotations
 
