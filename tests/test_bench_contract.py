@@ -86,3 +86,21 @@ def test_bench_torchrun_resnet50_two_ranks():
     lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
     out = _check_json(lines[0], 2)
     assert out["config"]["model"] == "resnet50"
+
+
+def test_bench_torchrun_eight_ranks():
+    """The full driver SCALE shape at N=8 (gloo, MLP) — rendezvous,
+    8-way bucket math, aggregate math."""
+    env = dict(os.environ)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port()), "bench.py", "--gpus", "8",
+         "--model", "mlp", "--steps", "2", "--warmup", "1",
+         "--device", "cpu"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + "\n" + r.stderr
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    out = _check_json(lines[0], 8)
+    assert out["config"]["parallelism"] == "dp8"
+    assert out["config"]["global_batch"] == 8 * 64
