@@ -121,6 +121,12 @@ def main():
         os.environ.setdefault("BPS_NUM_SERVER", "1")
         os.environ.setdefault("BPS_SERVER_URIS", "127.0.0.1:%d" % ps_port)
         os.environ.setdefault("BPS_MIN_COMPRESS_BYTES", "65536")
+        # colocated-PS sweet spot (same-box sweep, profiles/
+        # MEASUREMENTS.md): 32 MiB buckets halve the per-bucket server
+        # round-trip count; 8/16/64/104 MiB all measured slower
+        if not args.partition_mb:
+            os.environ.setdefault("BPS_PARTITION_BYTES",
+                                  str(32 * 2**20))
         if int(os.environ.get("LOCAL_RANK", "0")) == 0:
             from byteps_amd.ops import core
             server = core().Server(
