@@ -444,3 +444,48 @@ def test_ps_two_servers_parity():
     finally:
         s1.stop()
         s2.stop()
+
+
+# -- gradient accumulation through the PS path -------------------------------
+
+def _ps_accum_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = bps.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05),
+        named_parameters=m.named_parameters(),
+        backward_passes_per_step=2)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    for _ in range(steps):
+        for _ in range(2):
+            ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+            opt.step()
+        opt.zero_grad()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_ps_with_grad_accumulation(server):
+    """backward_passes_per_step=2 through the full KV round trip: the
+    accumulated (doubled) gradient must reach the server once per
+    window."""
+    m = _make_model()
+    opt = torch.optim.SGD(m.parameters(), lr=0.05)
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(2)]
+    ys = [torch.randn(8, 4) for _ in range(2)]
+    for _ in range(3):
+        opt.zero_grad()
+        loss = 2 * sum(((m(x) - y) ** 2).mean()
+                       for x, y in zip(xs, ys)) / 2
+        loss.backward()
+        opt.step()
+    expected = [p.detach().clone() for p in m.parameters()]
+    results = run_in_processes(_ps_accum_worker, 2, 3,
+                               extra_env=_ps_env(server))
+    for got, exp in zip(results[0], expected):
+        assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5)
