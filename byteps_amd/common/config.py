@@ -105,9 +105,16 @@ class Config:
         c.local_size = env_int("BPS_LOCAL_SIZE", "BYTEPS_LOCAL_SIZE", "LOCAL_WORLD_SIZE", default=1)
         c.worker_id = env_int("BPS_WORKER_ID", "DMLC_WORKER_ID", default=0)
 
+        ps_mode = c.num_servers > 0 or env_bool(
+            "BPS_FORCE_DISTRIBUTED", "BYTEPS_FORCE_DISTRIBUTED",
+            default=False)
         c.partition_bytes = env_int(
             "BPS_PARTITION_BYTES", "BYTEPS_PARTITION_BYTES",
-            default=DEFAULT_PARTITION_BYTES)
+            # PS mode: 32 MiB halves the per-bucket server round-trip
+            # count (same-box sweep, profiles/MEASUREMENTS.md); pure-RCCL
+            # stays at 16 MiB (sized for xGMI ring overlap granularity)
+            default=(32 * 1024 * 1024) if ps_mode
+            else DEFAULT_PARTITION_BYTES)
         # round up to a multiple of local_size * PAGE_SIZE so per-rank shards
         # of reduce-scatter stay page aligned (mirrors reference rounding,
         # common/global.cc:134-144)
