@@ -40,6 +40,9 @@ class _State:
         self.device: Optional[torch.device] = None
         self.owns_process_group = False
         self.kv = None           # KV client handle (PS mode)
+        self.key_rounds = {}     # pkey → last push round (survives the
+                                 # pipeline across suspend/resume so the
+                                 # server's version gate stays aligned)
         self.tracer = None
         self.base_master_port: Optional[int] = None
         self.resume_count = 0

@@ -247,6 +247,10 @@ class PSPipeline:
                     ("vanilla", "1", "true")
                 ki = _KeyInfo(pkey, server, nelem, compressor,
                               compressor is not None and ef)
+                # elastic resume: continue the round sequence — a fresh
+                # round=0 would let version-gated pulls be answered with
+                # the PREVIOUS session's merged data
+                ki.round = C._state.key_rounds.get(pkey, 0)
                 self.keys[bucket.plan.index] = ki
             return ki
 
@@ -353,6 +357,7 @@ class PSPipeline:
         st = self._staging_for(bucket, ki)
         comp = ki.compressor
         ki.round += 1
+        C._state.key_rounds[ki.key] = ki.round
         on_gpu = shard.is_cuda
         tracer = C._state.tracer
         step = ki.round - 1
@@ -576,6 +581,7 @@ class TensorPipeline:
                 nelem = tensor.numel()
                 server = st.assigner.assign(pkey, nelem * 4)
                 ki = _KeyInfo(pkey, server, nelem, None)
+                ki.round = C._state.key_rounds.get(pkey, 0)
                 self.keys[name] = ki
         if not ki.initialized:
             payload = struct.pack("<QII", ki.nelem, self.world, 0)
@@ -585,6 +591,7 @@ class TensorPipeline:
             self.kv.wait(t)
             ki.initialized = True
         ki.round += 1
+        C._state.key_rounds[ki.key] = ki.round
         src = tensor.detach()
         host = src.float().cpu().contiguous() if src.is_cuda \
             else src.float().contiguous()

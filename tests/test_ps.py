@@ -489,3 +489,41 @@ def test_ps_with_grad_accumulation(server):
                                extra_env=_ps_env(server))
     for got, exp in zip(results[0], expected):
         assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5)
+
+
+# -- elastic suspend/resume THROUGH the PS path ------------------------------
+
+def _ps_resume_worker(rank, world, steps):
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    opt = bps.DistributedOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.05),
+        named_parameters=m.named_parameters())
+    torch.manual_seed(42)
+    xs = [torch.randn(8, 16) for _ in range(world)]
+    ys = [torch.randn(8, 4) for _ in range(world)]
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        opt.step()
+    bps.suspend()
+    bps.resume(num_workers=world, num_servers=1)
+    for _ in range(steps):
+        opt.zero_grad()
+        ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+        opt.step()
+    out = [p.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return out
+
+
+def test_ps_suspend_resume_continuity(server):
+    """Suspend + resume with the SAME world keeps PS keys stable (server
+    re-init validation accepts the matching config) and training
+    continues exactly."""
+    expected = _baseline(1, 6)
+    results = run_in_processes(_ps_resume_worker, 1, 3,
+                               extra_env=_ps_env(server))
+    for got, exp in zip(results[0], expected):
+        assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
