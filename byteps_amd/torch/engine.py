@@ -184,6 +184,21 @@ class GradEngine:
                     b.buffer, dtype=self.comm_dtype)
         self._fused_desc = None     # (desc dev tensor, total_vec, vec)
 
+        # multi-ring collectives (reference BYTEPS_NCCL_NUM_RINGS /
+        # NcclManagerExpr, nccl_manager.cc:216-318): buckets round-robin
+        # across N communicators so RCCL can overlap their ring
+        # reductions instead of serializing on one comm's stream.
+        # Default 1 (RCCL's own pipelining is usually enough); the knob
+        # exists for the 8-GPU operator (docs/preflight-8gpu.md).
+        from ..common.config import env_int
+        nrings = max(1, env_int("BPS_NUM_RINGS", "BYTEPS_NCCL_NUM_RINGS",
+                                default=1))
+        self._rings: List = []
+        if (nrings > 1 and self.group is None and self.world > 1
+                and dist.is_initialized()):
+            ranks = list(range(self.world))
+            self._rings = [dist.new_group(ranks) for _ in range(nrings)]
+
         self._ps = None
         if C._state.ps_enabled:
             from . import ps_pipeline
@@ -398,14 +413,19 @@ class GradEngine:
                     self._wire_scratch[b.plan.index] = scratch
                 scratch.copy_(b.buffer)
                 b.work = dist.all_reduce(
-                    scratch, op=dist.ReduceOp.SUM, group=self.group,
-                    async_op=True)
+                    scratch, op=dist.ReduceOp.SUM,
+                    group=self._ring_for(b), async_op=True)
             else:
                 b.work = dist.all_reduce(
-                    b.buffer, op=dist.ReduceOp.SUM, group=self.group,
-                    async_op=True)
+                    b.buffer, op=dist.ReduceOp.SUM,
+                    group=self._ring_for(b), async_op=True)
         if self.on_bucket_issued is not None:
             self.on_bucket_issued(b)
+
+    def _ring_for(self, b: Bucket):
+        if not self._rings:
+            return self.group
+        return self._rings[b.plan.index % len(self._rings)]
 
     def _debug_sample(self, b: Bucket, stage: str) -> None:
         """BPS_DEBUG_SAMPLE_TENSOR: print first/last values of a watched

@@ -148,3 +148,16 @@ def test_cross_barrier_ps_world2():
                 assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5)
     finally:
         srv.stop()
+
+
+def test_cross_barrier_bf16_wire_world2():
+    """CrossBarrier with the reduced-precision wire (BPS_COMM_DTYPE=bf16):
+    per-bucket cast-back + averaging fused in _finish_bucket."""
+    expected = _baseline(2, 4)
+    results = run_in_processes(_cb_worker, 2, 4,
+                               extra_env={"BPS_COMM_DTYPE": "bf16"})
+    for r in range(2):
+        for got, exp in zip(results[r], expected):
+            # bf16 wire: ~3 decimal digits per step, 4 steps
+            assert torch.allclose(got, exp, rtol=5e-2, atol=5e-3), \
+                "bf16-wire cross-barrier diverged"

@@ -116,3 +116,31 @@ def test_ps_sharded_world(server, world):
         for got, exp in zip(results[r], expected):
             assert torch.allclose(got, exp, rtol=1e-4, atol=1e-5), \
                 "PS world=%d rank=%d param mismatch" % (world, r)
+
+
+def _rings_worker(rank, world, nrings):
+    import os
+    os.environ["BPS_NUM_RINGS"] = str(nrings)
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    from byteps_amd.torch.engine import GradEngine
+    eng = GradEngine(list(m.named_parameters()), partition_bytes=4096)
+    assert len(eng._rings) == nrings
+    xs, ys = _data(world)
+    ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+    eng.synchronize()
+    grads = [p.grad.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return grads
+
+
+def test_multi_ring_collectives_world4():
+    """BPS_NUM_RINGS=2: buckets round-robin across two communicators;
+    results must match the single-ring big-batch baseline (reference
+    NcclManagerExpr multi-ring, nccl_manager.cc:216-318)."""
+    expected = _baseline_grads(4)
+    results = run_in_processes(_rings_worker, 4, 2)
+    for r in range(4):
+        for got, exp in zip(results[r], expected):
+            assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
