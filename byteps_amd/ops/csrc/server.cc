@@ -275,6 +275,7 @@ struct KeyState {
   std::vector<float> scratch;    // decompress workspace
   std::vector<int8_t> code_scratch;  // dithering dense-code workspace
   bool server_ef = false;        // error-feedback on the merged reply
+  bool dense_reply = false;      // colocated worker: skip Elias coding
   std::vector<float> ef_err;     // residual of the previous reply
   std::vector<float> ef_comp;    // compensated merge workspace
 };
@@ -626,6 +627,7 @@ class Server {
           slot->levels = std::max(1u, ip.levels);
           slot->async_mode = cmd_async(h.cmd);
           slot->server_ef = (ip.flags & 1u) != 0;
+          slot->dense_reply = (ip.flags & 2u) != 0;
         }
       }
       if (!slot) {
@@ -636,6 +638,7 @@ class Server {
         slot->levels = std::max(1u, ip.levels);
         slot->async_mode = cmd_async(h.cmd);
         slot->server_ef = (ip.flags & 1u) != 0;
+        slot->dense_reply = (ip.flags & 2u) != 0;
         // store allocation is DEFERRED to the first push so the
         // engine thread that owns this key first-touches the pages on
         // its own NUMA node (reference pre-allocated page-aligned at
@@ -1041,7 +1044,8 @@ class Server {
         ks->reply.resize(5 + n);
         std::memcpy(ks->reply.data(), &norm, 4);
         int64_t wlen = 0;
-        if (bps_cpu_dither_encode(ks->code_scratch.data(), n,
+        if (!ks->dense_reply &&
+            bps_cpu_dither_encode(ks->code_scratch.data(), n,
                                   (uint8_t*)ks->reply.data() + 5, n,
                                   &wlen) == 0) {
           ks->reply[4] = 1;          // Elias sparse wire

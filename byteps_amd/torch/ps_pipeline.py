@@ -281,7 +281,7 @@ class PSPipeline:
                 self._staging[bucket.plan.index] = s
             return s
 
-    def _ensure_init(self, ki: _KeyInfo) -> None:
+    def _ensure_init(self, ki: _KeyInfo, st=None) -> None:
         if ki.initialized:
             return
         comp = ki.compressor
@@ -289,6 +289,10 @@ class PSPipeline:
         levels = comp.levels if comp is not None else 0
         # bit0: ask the server to run error feedback on its merged reply
         flags = 1 if ki.server_ef else 0
+        # bit1: colocated shm lane — skip bit-level wire coding on BOTH
+        # sides (Elias compression buys nothing when no NIC is crossed)
+        if st is not None and st.ipc:
+            flags |= 2
         payload = struct.pack("<QIIII", ki.nelem, self.num_nodes, levels,
                               flags, 0)
         buf = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
@@ -353,8 +357,8 @@ class PSPipeline:
                     rs_event) -> tuple:
         """Runs in a pool thread: compress → D2H → push/pull → H2D →
         decompress.  Returns (done_event | None, reply_fp32 | None)."""
-        self._ensure_init(ki)
         st = self._staging_for(bucket, ki)
+        self._ensure_init(ki, st)
         comp = ki.compressor
         ki.round += 1
         C._state.key_rounds[ki.key] = ki.round
@@ -394,7 +398,8 @@ class PSPipeline:
         # dense staging into the wire buffer; fall back to dense when the
         # stream would be larger
         push_buf = st.send
-        if comp is not None and getattr(comp, "host_wire", False):
+        if comp is not None and getattr(comp, "host_wire", False) \
+                and not st.ipc:
             wlen = comp.encode_wire(st.send[:nbytes], ki.nelem, st.wire)
             if wlen > 0:
                 push_buf = st.wire

@@ -177,7 +177,10 @@ def test_ps_dithering_sparse_wire_end_to_end():
             env = {"BPS_FORCE_DISTRIBUTED": "1",
                    "BPS_SERVER_URIS": "127.0.0.1:%d" % srv.port,
                    "BPS_NUM_SERVER": "1",
-                   "BPS_MIN_COMPRESS_BYTES": "0"}
+                   "BPS_MIN_COMPRESS_BYTES": "0",
+                   # force the TCP lane so the Elias wire actually
+                   # engages (colocated shm skips bit-level coding)
+                   "BPS_ENABLE_IPC": "0"}
             results[sparse] = run_in_processes(_dither_ps_worker, 1,
                                                sparse, extra_env=env)[0]
         finally:
