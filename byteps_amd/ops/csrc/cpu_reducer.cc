@@ -298,6 +298,21 @@ int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
   return 0;
 }
 
+// fused decode → accumulator (one pass, like the onebit/dithering paths)
+int bps_cpu_fp8_accumulate(const uint8_t* code, int64_t n, float amax,
+                           float* acc, int first) {
+  const float inv = amax / 448.0f;
+  if (first) {
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i) acc[i] = fp8_e4m3_decode(code[i]) * inv;
+  } else {
+#pragma omp parallel for
+    for (int64_t i = 0; i < n; ++i)
+      acc[i] += fp8_e4m3_decode(code[i]) * inv;
+  }
+  return 0;
+}
+
 // ---------------------------------------------------------------------------
 // Elias-delta sparse wire for dithering codes (reference
 // common/compressor/utils.h:115-250 BitWriter/Elias-delta +

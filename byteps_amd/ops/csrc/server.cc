@@ -95,6 +95,8 @@ int bps_cpu_topk_select(const float* x, int64_t n, int64_t k, int32_t* idx,
                         float* val);
 int bps_cpu_fp8_compress(const float* x, int64_t n, float amax,
                          uint8_t* code);
+int bps_cpu_fp8_accumulate(const uint8_t* code, int64_t n, float amax,
+                           float* acc, int first);
 int bps_cpu_fp8_decompress(const uint8_t* code, int64_t n, float amax,
                            float* out);
 float bps_cpu_norm(const float* x, int64_t n, int mode);
