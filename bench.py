@@ -46,6 +46,11 @@ def parse_args():
                    choices=["none", "onebit", "topk", "randomk", "dithering", "fp8"])
     p.add_argument("--partition-mb", type=int, default=0,
                    help="override BPS_PARTITION_BYTES (MiB)")
+    p.add_argument("--cross-barrier", action="store_true",
+                   help="pipelined per-bucket updates (reference "
+                        "cross_barrier.py): optimizer applies as each "
+                        "bucket's communication lands; the next forward "
+                        "blocks per-layer on its params")
     return p.parse_args()
 
 
@@ -157,10 +162,19 @@ def main():
             # Elias wire)
             cparams["compressor_k"] = 8
             cparams["partition"] = "natural"
-    from byteps_amd.torch.parallel import DistributedDataParallel as DDP
-    model = DDP(net, broadcast_buffers=False, compression_params=cparams)
-    opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
-                          weight_decay=1e-4)
+    cb = None
+    if args.cross_barrier:
+        from byteps_amd.torch.cross_barrier import CrossBarrier
+        opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
+                              weight_decay=1e-4)
+        cb = CrossBarrier(net, opt)
+        model = net
+    else:
+        from byteps_amd.torch.parallel import DistributedDataParallel as DDP
+        model = DDP(net, broadcast_buffers=False,
+                    compression_params=cparams)
+        opt = torch.optim.SGD(net.parameters(), lr=0.1, momentum=0.9,
+                              weight_decay=1e-4)
 
     # pre-tuned hipBLASLt algorithm selections for gfx950 (TunableOp,
     # generated offline by scripts/gen_tunableop.py — tuning is too slow
@@ -182,6 +196,13 @@ def main():
         else torch.autocast("cpu", enabled=False)
 
     def train_step():
+        if cb is not None:
+            cb.zero_grad()
+            with autocast:
+                loss = step_fn(model)
+            loss.backward()
+            cb.step()             # returns immediately; updates pipeline
+            return loss
         model.zero_grad_buckets()
         with autocast:
             loss = step_fn(model)
@@ -196,8 +217,8 @@ def main():
     # gain measured ~0.3% against real replay risk on an untested topology
     want_graph = args.graph == "on" or (
         args.graph == "auto" and on_gpu and world == 1)
-    if args.compression != "none":
-        want_graph = False      # PS pipeline does host-side KV work per step
+    if args.compression != "none" or args.cross_barrier:
+        want_graph = False      # host-side KV / poller work per step
     if want_graph and on_gpu:
         try:
             for _ in range(3):      # warm up allocator + RCCL before capture
@@ -239,6 +260,8 @@ def main():
     t0 = time.perf_counter()
     for _ in range(args.steps):
         run_step()
+    if cb is not None:
+        cb.synchronize()          # drain in-flight per-bucket updates
     if on_gpu:
         torch.cuda.synchronize()
     t1 = time.perf_counter()
@@ -277,9 +300,12 @@ def main():
                 "parallelism": "dp%d" % world,
                 "graph": graphed is not None,
                 "compression": args.compression,
+                "cross_barrier": bool(args.cross_barrier),
             },
         }
         print(json.dumps(out))
+    if cb is not None:
+        cb.stop()
     bps.shutdown()
     if server is not None:
         server.stop()

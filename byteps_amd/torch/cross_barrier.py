@@ -42,9 +42,17 @@ class CrossBarrier:
 
     def __init__(self, model: torch.nn.Module, optimizer,
                  named_parameters=None, process_group=None,
-                 partition_bytes: Optional[int] = None):
+                 partition_bytes: Optional[int] = None,
+                 auto_zero: bool = True):
         self.model = model
         self.optimizer = optimizer
+        # auto_zero: the poller zeroes each bucket right after its
+        # update lands, so zero_grad() need not drain the pipeline and
+        # step N+1's backward overlaps step N's tail (the point of the
+        # reference's cross-barrier).  Buckets holding SPLIT params are
+        # excluded (their views feed the split copy-back) and zeroed at
+        # split-completion instead.
+        self.auto_zero = auto_zero
         if named_parameters is None:
             named_parameters = model.named_parameters()
         named = [(n, p) for n, p in named_parameters if p.requires_grad]
@@ -166,6 +174,13 @@ class CrossBarrier:
         if opt is not None:
             self._refresh_hyper(opt)
             opt.step()
+        if self.auto_zero:
+            # zero per-span: split params' views are WRITTEN (not
+            # accumulated) at hook time and still feed the split
+            # copy-back, so only non-split spans reset here
+            for q, g in zip(bucket.params, bucket.grads):
+                if id(q) not in self._split_pidx_of:
+                    g.zero_()
         # split params: when the last containing bucket lands, refresh
         # p.grad from the averaged shard views and step exactly once
         for p in bucket.params:
@@ -186,6 +201,9 @@ class CrossBarrier:
             sopt = self._split_opts[pidx]
             self._refresh_hyper(sopt)
             sopt.step()
+            if self.auto_zero:
+                p.grad.zero_()      # autograd accumulates into the
+                                    # private grad; views are copy-target
             self._events[id(p)].set()
         bucket.reset()
 
@@ -218,6 +236,8 @@ class CrossBarrier:
     # -- user API -----------------------------------------------------------
 
     def zero_grad(self) -> None:
+        if self.auto_zero:
+            return                  # poller zeroes per bucket at update
         self.synchronize()          # grads must not be zeroed mid-update
         self._engine.zero_grad()
 
