@@ -894,12 +894,8 @@ class Server {
         float amax;
         std::memcpy(&amax, t.data(), 4);
         const uint8_t* code = (const uint8_t*)(t.data() + 4);
-        ks->scratch.resize(n);
-        bps_cpu_fp8_decompress(code, n, amax, ks->scratch.data());
-        if (first)
-          std::memcpy(acc, ks->scratch.data(), n * sizeof(float));
-        else
-          bps_cpu_sum(acc, ks->scratch.data(), n, 0);
+        // fused decode→accumulate, one pass
+        bps_cpu_fp8_accumulate(code, n, amax, acc, first ? 1 : 0);
         break;
       }
       default:
