@@ -122,9 +122,14 @@ def _grow_worker(rank, world, tmpdir):
     def mark(tag):
         open(os.path.join(tmpdir, "%s.%d" % (tag, rank)), "w").close()
 
-    def wait(tag, ranks):
+    def wait(tag, ranks, deadline=180.0):
+        t0 = time.time()
         while not all(os.path.exists(os.path.join(tmpdir, "%s.%d" % (tag, r)))
                       for r in ranks):
+            if time.time() - t0 > deadline:
+                raise RuntimeError(
+                    "rank %d: timed out waiting for %s markers (a peer "
+                    "died before marking — see its traceback)" % (rank, tag))
             time.sleep(0.05)
 
     if rank < 2:
