@@ -32,6 +32,7 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> st
     if not force and not _newer_than_out(srcs + hdrs):
         return OUT
     hipcc = os.environ.get("HIPCC", "hipcc")
+    tmp_out = OUT + ".tmp"
     cmd = [
         hipcc, "--offload-arch=" + arch, "-O3", "-std=c++17", "-fPIC",
         "-shared", "-fopenmp", "-pthread", "-fvisibility=hidden",
@@ -40,11 +41,14 @@ def build(force: bool = False, arch: str = "gfx950", verbose: bool = True) -> st
         "-I", pybind11.get_include(),
         "-I", sysconfig.get_paths()["include"],
         *srcs,
-        "-o", OUT,
+        "-o", tmp_out,
     ]
     if verbose:
         print("[byteps_amd build]", " ".join(cmd), file=sys.stderr)
     subprocess.run(cmd, check=True)
+    # atomic swap: a concurrently-spawning process must never import a
+    # half-written .so
+    os.replace(tmp_out, OUT)
     return OUT
 
 
