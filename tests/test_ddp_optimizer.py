@@ -145,3 +145,37 @@ def test_broadcast_optimizer_state():
     assert len(s0) == len(s1)
     for a, b in zip(s0, s1):
         assert float(a) == float(b)
+
+
+# -- buffer broadcast (BN running stats sync each forward) -------------------
+
+def _buffers_worker(rank, world):
+    import byteps_amd.torch as bps
+    from byteps_amd.torch.parallel import DistributedDataParallel as DDP
+    bps.init()
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(torch.nn.Linear(8, 8),
+                            torch.nn.BatchNorm1d(8))
+    net = DDP(m, broadcast_buffers=True)
+    # desync buffers on non-root, then one forward must re-sync them
+    if rank != 0:
+        with torch.no_grad():
+            m[1].running_mean.fill_(42.0)
+    torch.manual_seed(100 + rank)
+    net(torch.randn(4, 8))      # train forward: desync is re-broadcast,
+                                # then BN re-updates stats per-rank
+    m.eval()                    # eval: no stat update after the sync
+    net(torch.randn(4, 8))
+    out = [b.detach().clone() for b in m.buffers()]
+    bps.shutdown()
+    return out
+
+
+def test_ddp_broadcast_buffers_each_forward():
+    """DDP re-broadcasts buffers from rank 0 before each forward
+    (reference parallel/distributed.py:209-220) — a desynced running
+    stat on rank 1 must be overwritten."""
+    results = run_in_processes(_buffers_worker, 2)
+    for b0, b1 in zip(results[0], results[1]):
+        assert torch.equal(b0, b1), "buffers diverged across ranks"
+    assert not torch.any(results[1][0] == 42.0), "rank-1 buffer not resynced"
