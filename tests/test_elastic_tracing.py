@@ -195,3 +195,20 @@ def test_elastic_rebucket_grow_2_to_3(tmp_path):
     for got, exp in zip(results[0][0], [p.grad for p in m.parameters()]):
         assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6), \
             "averaged grads diverge from the 3-way big batch"
+
+
+# -- dist_launch units (SSH fan-out command construction) --------------------
+
+def test_dist_launch_hostfile_and_env(tmp_path, monkeypatch):
+    from byteps_amd.launcher.dist_launch import read_hostfile, forwarded_env
+    hf = tmp_path / "hosts"
+    hf.write_text("10.0.0.1:8\n# comment\n10.0.0.2\n\n10.0.0.3:4 # gpu4\n")
+    assert read_hostfile(str(hf)) == ["10.0.0.1", "10.0.0.2", "10.0.0.3"]
+    monkeypatch.setenv("BPS_NUM_SERVER", "2")
+    monkeypatch.setenv("DMLC_PS_ROOT_URI", "10.0.0.9")
+    monkeypatch.setenv("HOME", "/root")          # must NOT forward
+    pairs = forwarded_env(["EXTRA=1"])
+    assert "BPS_NUM_SERVER=2" in pairs
+    assert "DMLC_PS_ROOT_URI=10.0.0.9" in pairs
+    assert "EXTRA=1" in pairs
+    assert not any(p.startswith("HOME=") for p in pairs)
