@@ -527,3 +527,55 @@ def test_ps_suspend_resume_continuity(server):
                                extra_env=_ps_env(server))
     for got, exp in zip(results[0], expected):
         assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
+
+
+def test_server_process_entry():
+    """`python -m byteps_amd.server` — the standalone server process
+    (reference byteps/server/__init__.py) — serves a KV round trip."""
+    import os
+    import re
+    import signal
+    import struct
+    import subprocess
+    import sys
+    import time
+    env = dict(os.environ)
+    env["BPS_SERVER_PORT"] = "0"
+    env["BPS_LOG_LEVEL"] = "INFO"
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "byteps_amd.server"],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True, cwd=os.path.dirname(os.path.dirname(
+            os.path.abspath(__file__))))
+    try:
+        port = None
+        t0 = time.time()
+        while time.time() - t0 < 30:
+            line = proc.stdout.readline()
+            m = re.search(r"listening on :(\d+)", line or "")
+            if m:
+                port = int(m.group(1))
+                break
+        assert port, "server did not announce its port"
+        from byteps_amd.ops import _core
+        kv = _core.KVClient(0, ["127.0.0.1:%d" % port])
+        payload = struct.pack("<QIIII", 128, 1, 0, 0, 0)
+        pb = torch.frombuffer(bytearray(payload), dtype=torch.uint8)
+        t = kv.submit(0, 5, 3, pb.data_ptr(), len(payload), 0, 0, 0, 0)
+        kv.wait(t)
+        x = torch.randn(128)
+        send = x.view(torch.uint8).reshape(-1).clone()
+        t = kv.submit(0, 1, 3, send.data_ptr(), 512, 0, 0, 0, 1)
+        kv.wait(t)
+        recv = torch.empty(512, dtype=torch.uint8)
+        t = kv.submit(0, 2, 3, 0, 0, recv.data_ptr(), 512, 0, 1)
+        rl, _ = kv.wait(t)
+        assert rl == 512
+        assert torch.equal(recv.view(torch.float32), x)
+        kv.close()
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=10)
+        except subprocess.TimeoutExpired:
+            proc.kill()
