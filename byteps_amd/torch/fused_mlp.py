@@ -5,8 +5,9 @@ The eager path pays a standalone GELU-backward kernel plus two bias-grad
 reductions per layer (~2 ms/step on BERT-large,
 profiles/bert_large_final_kernels.txt).  This hipBLASLt build for gfx950
 supports DGELU (NN) and BGRADB (NT) epilogues but no AUX-output forward
-(probed — scripts/probe_blaslt.cc), so the FORWARD stays eager (addmm +
-gelu already keeps the pre-GELU activation) and the BACKWARD is fused:
+(probed — scripts/probe_blaslt.cc), so the FORWARD stays on torch's
+TunableOp-tuned addmm + gelu (keeping the pre-GELU activation) and the
+BACKWARD goes through hipBLASLt with fused bias-grads:
 
   dgrad2: dY1 = gelu_backward(dY2·W2, H)   (lt GEMM + aten kernel —
           the DGELU epilogue exists but measured 4x slower than the
