@@ -144,3 +144,32 @@ def test_multi_ring_collectives_world4():
     for r in range(4):
         for got, exp in zip(results[r], expected):
             assert torch.allclose(got, exp, rtol=1e-5, atol=1e-6)
+
+
+def _rings_wire_worker(rank, world):
+    import os
+    os.environ["BPS_NUM_RINGS"] = "2"
+    os.environ["BPS_COMM_DTYPE"] = "bf16"
+    import byteps_amd.torch as bps
+    bps.init()
+    m = _make_model()
+    from byteps_amd.torch.engine import GradEngine
+    eng = GradEngine(list(m.named_parameters()), partition_bytes=4096)
+    xs, ys = _data(world)
+    ((m(xs[rank]) - ys[rank]) ** 2).mean().backward()
+    eng.synchronize()
+    grads = [p.grad.detach().clone() for p in m.parameters()]
+    bps.shutdown()
+    return grads
+
+
+def test_rings_with_bf16_wire_world2():
+    """Multi-ring + reduced-precision wire: scratch all-reduces spread
+    across rings, fused cast-back epilogue on the CPU fallback path."""
+    expected = _baseline_grads(2)
+    results = run_in_processes(_rings_wire_worker, 2)
+    for r in range(2):
+        for got, exp in zip(results[r], expected):
+            assert torch.allclose(got, exp, rtol=3e-2, atol=3e-3)
+        for a, b in zip(results[0], results[1]):
+            assert torch.equal(a, b)
